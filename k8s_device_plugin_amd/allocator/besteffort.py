@@ -1,0 +1,195 @@
+"""Best-effort topology-aware preferred allocation.
+
+Semantics match the reference policy (reference:
+internal/pkg/allocator/besteffort_policy.go:88-151, device.go:255-443):
+
+  - fast paths: available==size -> available; required==size -> required;
+  - partitions are grouped by physical GPU (devID); groups are sorted
+    ascending by free-partition count (tie: parent id) so nearly-full GPUs
+    are packed first (anti-fragmentation);
+  - candidate sets are seeded per group (prefer one whole GPU's partitions)
+    and breadth-first extended across groups when one GPU cannot satisfy
+    the request;
+  - the candidate with the minimum total pairwise weight wins.
+
+The pair weights themselves are hive-aware (weights.py), which on an
+8*MI355X node packs 2/4/8-GPU requests onto one xGMI hive.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, Iterable, List, Optional, Sequence
+
+from ..topology.discovery import GPUDevice
+from ..topology.kfd import KFDTopology
+from ..topology.sysfs import SysPaths
+from .weights import compute_pair_weights
+
+
+class AllocationError(ValueError):
+    pass
+
+
+@dataclass
+class _Group:
+    """Free partitions of one physical GPU available to this request."""
+
+    dev_id: str
+    parent_id: str
+    node_ids: List[int]  # ascending
+
+
+@dataclass
+class _Subset:
+    ids: List[int]
+    parents: frozenset
+    weight: int
+
+    def extended(self, node_id: int, parent_idx: int,
+                 weights: Dict[int, Dict[int, int]]) -> "_Subset":
+        w = self.weight
+        for other in self.ids:
+            frm, to = (other, node_id) if other < node_id else (node_id, other)
+            w += weights.get(frm, {}).get(to, 0)
+        return _Subset(self.ids + [node_id], self.parents | {parent_idx}, w)
+
+
+class BestEffortPolicy:
+    """Preferred-allocation policy; init once at plugin start, then allocate
+    from memory with zero I/O (reference: plugin.go:85)."""
+
+    def __init__(self) -> None:
+        self._devices: Dict[str, GPUDevice] = {}
+        self._weights: Dict[int, Dict[int, int]] = {}
+        self._groups: Dict[str, _Group] = {}
+
+    def init(
+        self,
+        devices: Iterable[GPUDevice],
+        topology: Optional[KFDTopology] = None,
+        paths: SysPaths = SysPaths(),
+    ) -> None:
+        devs = list(devices)
+        if not devs:
+            raise AllocationError("no devices to initialize allocator with")
+        topo = topology if topology is not None else KFDTopology.load(paths)
+        self._weights = compute_pair_weights(devs, topo)
+        if not self._weights:
+            raise AllocationError("failed to initialize pair weights (no device links)")
+        self._devices = {d.id: d for d in devs}
+        self._groups = {}
+        for d in devs:
+            g = self._groups.setdefault(d.dev_id, _Group(d.dev_id, "", []))
+            g.node_ids.append(d.node_id)
+            if not d.is_partition:
+                g.parent_id = d.id
+        for g in self._groups.values():
+            g.node_ids.sort()
+
+    @property
+    def initialized(self) -> bool:
+        return bool(self._weights)
+
+    def allocate(
+        self,
+        available_ids: Sequence[str],
+        required_ids: Sequence[str],
+        size: int,
+    ) -> List[str]:
+        if size <= 0:
+            raise AllocationError("allocation size must be a positive integer")
+        if len(available_ids) < size:
+            raise AllocationError("available devices count less than allocation size")
+        if len(required_ids) > size:
+            raise AllocationError("must-include set larger than allocation size")
+        if len(required_ids) > len(available_ids):
+            raise AllocationError("must-include set larger than available set")
+        if not self._devices:
+            raise AllocationError("allocator not initialized")
+        if len(available_ids) == size:
+            return list(available_ids)
+        if len(required_ids) == size:
+            return list(required_ids)
+        if not set(required_ids).issubset(set(available_ids)):
+            raise AllocationError("must-include devices not all available")
+
+        available = [self._devices[i] for i in available_ids if i in self._devices]
+        required = [self._devices[i] for i in required_ids if i in self._devices]
+        candidates = self._candidate_subsets(available, required, size)
+        if not candidates:
+            raise AllocationError("no candidate subset found with matching criteria")
+
+        best = min(candidates, key=lambda s: s.weight)
+        by_node = {d.node_id: d.id for d in available}
+        return [by_node[nid] for nid in best.ids if nid in by_node]
+
+    # ---- internals ----
+
+    def _filtered_groups(
+        self, available: Sequence[GPUDevice], required: Sequence[GPUDevice]
+    ) -> List[_Group]:
+        avail_ids = {d.node_id for d in available}
+        req_ids = {d.node_id for d in required}
+        groups: List[_Group] = []
+        for g in self._groups.values():
+            ids = sorted(i for i in g.node_ids if i in avail_ids and i not in req_ids)
+            if ids:
+                groups.append(_Group(g.dev_id, g.parent_id, ids))
+        groups.sort(key=lambda g: (len(g.node_ids), g.parent_id))
+        return groups
+
+    def _candidate_subsets(
+        self,
+        available: Sequence[GPUDevice],
+        required: Sequence[GPUDevice],
+        size: int,
+    ) -> List[_Subset]:
+        groups = self._filtered_groups(available, required)
+        new_size = size - len(required)
+        req_node_ids = [d.node_id for d in required]
+
+        def finish(s: _Subset) -> _Subset:
+            for rid in req_node_ids:
+                s = s.extended(rid, -1, self._weights)
+            return s
+
+        final: List[_Subset] = []
+        queue: List[_Subset] = []
+
+        for idx, g in enumerate(groups):
+            s = _Subset([g.node_ids[0]], frozenset([idx]), 0)
+            if new_size == 1:
+                final.append(finish(s))
+                continue
+            fulfilled = False
+            for nid in g.node_ids[1:]:
+                s = s.extended(nid, idx, self._weights)
+                if len(s.ids) == new_size:
+                    fulfilled = True
+                    break
+            if fulfilled:
+                final.append(finish(s))
+            else:
+                queue.append(s)
+
+        # breadth-first extension across GPUs for requests no single GPU
+        # can satisfy (reference: device.go:406-441)
+        while queue:
+            cur = queue.pop(0)
+            if len(cur.parents) == len(groups):
+                continue
+            for idx, g in enumerate(groups):
+                if idx in cur.parents:
+                    continue
+                s = _Subset(cur.ids, cur.parents | {idx}, cur.weight)
+                done = False
+                for nid in g.node_ids:
+                    s = s.extended(nid, idx, self._weights)
+                    if len(s.ids) == new_size:
+                        final.append(finish(s))
+                        done = True
+                        break
+                if not done:
+                    queue.append(s)
+        return final

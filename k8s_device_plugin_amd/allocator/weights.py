@@ -1,0 +1,94 @@
+"""Pairwise GPU connectivity weights (lower = closer).
+
+Base model is compatible with the reference's scoring (reference:
+internal/pkg/allocator/device.go:38-55,136-158): same physical GPU +10 /
+different +20; xGMI link +10, PCIe +40, other +50; same NUMA +10 /
+different +20.
+
+MI355X-first extension: kfd exposes `hive_id` for xGMI hives; a pair in
+DIFFERENT hives gets a +30 penalty so multi-GPU requests pack onto one hive
+and downstream RCCL ring all-reduce runs on the 7x ~153 GB/s point-to-point
+xGMI links rather than crossing hives/PCIe.  Pairs with
+unknown hive (hive_id 0) are unaffected, which keeps scoring identical to the
+reference on topologies without hive information.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Dict, Iterable, List, Optional
+
+from ..topology.discovery import GPUDevice
+from ..topology.kfd import KFDLink, KFDTopology
+
+SAME_DEV_WEIGHT = 10
+DIFFERENT_DEV_WEIGHT = 20
+XGMI_LINK_WEIGHT = 10
+PCIE_LINK_WEIGHT = 40
+OTHER_LINK_WEIGHT = 50
+SAME_NUMA_WEIGHT = 10
+DIFFERENT_NUMA_WEIGHT = 20
+CROSS_HIVE_WEIGHT = 30
+
+PAIR_WEIGHTS = {
+    "same_dev": SAME_DEV_WEIGHT,
+    "different_dev": DIFFERENT_DEV_WEIGHT,
+    "xgmi": XGMI_LINK_WEIGHT,
+    "pcie": PCIE_LINK_WEIGHT,
+    "other_link": OTHER_LINK_WEIGHT,
+    "same_numa": SAME_NUMA_WEIGHT,
+    "different_numa": DIFFERENT_NUMA_WEIGHT,
+    "cross_hive": CROSS_HIVE_WEIGHT,
+}
+
+
+def pair_weight(
+    from_dev: GPUDevice,
+    to_dev: GPUDevice,
+    link_type: int,
+    from_hive: int = 0,
+    to_hive: int = 0,
+) -> int:
+    w = SAME_DEV_WEIGHT if from_dev.dev_id == to_dev.dev_id else DIFFERENT_DEV_WEIGHT
+    if link_type == KFDLink.XGMI:
+        w += XGMI_LINK_WEIGHT
+    elif link_type == KFDLink.PCIE:
+        w += PCIE_LINK_WEIGHT
+    else:
+        w += OTHER_LINK_WEIGHT
+    w += (
+        SAME_NUMA_WEIGHT
+        if from_dev.numa_node == to_dev.numa_node
+        else DIFFERENT_NUMA_WEIGHT
+    )
+    if from_hive and to_hive and from_hive != to_hive:
+        w += CROSS_HIVE_WEIGHT
+    return w
+
+
+def compute_pair_weights(
+    devices: Iterable[GPUDevice], topology: KFDTopology
+) -> Dict[int, Dict[int, int]]:
+    """{min_node_id: {max_node_id: weight}} for every linked device pair.
+
+    Scans io_links + p2p_links of every valid GPU topology node, keeping only
+    pairs where both endpoints are schedulable devices (reference:
+    device.go:160-253).  Missing pairs score 0 when summed, matching the
+    reference's map-default semantics.
+    """
+    by_node: Dict[int, GPUDevice] = {d.node_id: d for d in devices}
+    weights: Dict[int, Dict[int, int]] = {}
+    for node in topology.nodes.values():
+        if node.render_minor <= 0:
+            continue
+        for link in node.all_links():
+            a, b = link.node_from, link.node_to
+            frm, to = (a, b) if a < b else (b, a)
+            if frm not in by_node or to not in by_node:
+                continue
+            from_hive = topology.nodes[frm].hive_id if frm in topology.nodes else 0
+            to_hive = topology.nodes[to].hive_id if to in topology.nodes else 0
+            weights.setdefault(frm, {})[to] = pair_weight(
+                by_node[frm], by_node[to], link.type, from_hive, to_hive
+            )
+    return weights

@@ -1,0 +1,175 @@
+"""Best-effort allocator tests.
+
+Scenario coverage mirrors the reference's table-driven policy tests
+(reference: internal/pkg/allocator/besteffort_policy_test.go,
+device_test.go) on equivalent topologies synthesized by fakesysfs:
+same-GPU packing, NUMA affinity, anti-fragmentation, required-IDs — plus the
+MI355X-native xGMI-hive packing the reference lacks.
+"""
+
+import pytest
+
+from k8s_device_plugin_amd.allocator import AllocationError, BestEffortPolicy
+from k8s_device_plugin_amd.allocator.weights import compute_pair_weights
+from k8s_device_plugin_amd.topology import KFDTopology, discover_gpus
+from k8s_device_plugin_amd.testing.fakesysfs import build_mi355x_node, FakeSysfs
+
+
+def make_policy(fs):
+    topo = KFDTopology.load(fs.paths)
+    devices = discover_gpus(fs.paths, topology=topo)
+    policy = BestEffortPolicy()
+    policy.init(devices.values(), topology=topo)
+    return policy, devices
+
+
+# ---------- 8x whole GPUs, one hive ----------
+
+def test_allocate_guards(fake_mi355x_8):
+    policy, devices = make_policy(fake_mi355x_8)
+    ids = sorted(devices)
+    with pytest.raises(AllocationError):
+        policy.allocate(ids, [], 0)
+    with pytest.raises(AllocationError):
+        policy.allocate(ids[:2], [], 3)
+    with pytest.raises(AllocationError):
+        policy.allocate(ids, ids[:3], 2)
+    with pytest.raises(AllocationError):
+        policy.allocate(ids[:4], ["not-a-dev"], 2)
+
+
+def test_allocate_fast_paths(fake_mi355x_8):
+    policy, devices = make_policy(fake_mi355x_8)
+    ids = sorted(devices)
+    assert policy.allocate(ids[:3], [], 3) == ids[:3]
+    assert policy.allocate(ids, ids[:2], 2) == ids[:2]
+
+
+def test_allocate_one(fake_mi355x_8):
+    policy, devices = make_policy(fake_mi355x_8)
+    ids = sorted(devices)
+    out = policy.allocate(ids, [], 1)
+    assert len(out) == 1 and out[0] in devices
+
+
+def test_numa_affinity(fake_mi355x_8):
+    """From GPUs 2..7 (numa0: 2,3; numa1: 4..7), a 4-GPU request must land
+    entirely on NUMA 1 (cf. reference besteffort_policy_test.go:91-96)."""
+    policy, devices = make_policy(fake_mi355x_8)
+    ids = sorted(devices)  # pci addresses sort by bus => GPU order
+    available = ids[2:]
+    out = policy.allocate(available, [], 4)
+    numas = {devices[i].numa_node for i in out}
+    assert numas == {1}, f"expected all-numa1 packing, got {out}"
+
+
+def test_required_kept_and_extended(fake_mi355x_8):
+    policy, devices = make_policy(fake_mi355x_8)
+    ids = sorted(devices)
+    # require one numa-0 GPU; best completion keeps numa-0 peers
+    out = policy.allocate(ids, [ids[1]], 2)
+    assert ids[1] in out and len(out) == 2
+    other = devices[[i for i in out if i != ids[1]][0]]
+    assert other.numa_node == devices[ids[1]].numa_node
+
+
+# ---------- CPX fan-out: 64 partitions ----------
+
+def test_cpx_pack_one_gpu(fake_mi355x_cpx):
+    """An 8-partition request must return one whole GPU's partitions
+    (cf. reference besteffort_policy_test.go:136-139)."""
+    policy, devices = make_policy(fake_mi355x_cpx)
+    ids = sorted(devices)
+    out = policy.allocate(ids, [], 8)
+    assert len(out) == 8
+    dev_ids = {devices[i].dev_id for i in out}
+    assert len(dev_ids) == 1, f"expected one-GPU packing, got {out}"
+
+
+def test_cpx_anti_fragmentation(fake_mi355x_cpx):
+    """With GPU A holding 3 free partitions and GPU B 8, a 3-partition
+    request must drain GPU A (most-used first, reference device.go:343-351)."""
+    policy, devices = make_policy(fake_mi355x_cpx)
+    by_gpu = {}
+    for d in devices.values():
+        by_gpu.setdefault(d.dev_id, []).append(d.id)
+    gpu_a, gpu_b = sorted(by_gpu)[:2]
+    available = sorted(by_gpu[gpu_a])[:3] + sorted(by_gpu[gpu_b])
+    out = policy.allocate(available, [], 3)
+    assert {devices[i].dev_id for i in out} == {gpu_a}
+
+
+def test_cpx_spillover_two_gpus(fake_mi355x_cpx):
+    """A 10-partition request spans exactly two GPUs (8 + 2)."""
+    policy, devices = make_policy(fake_mi355x_cpx)
+    ids = sorted(devices)
+    out = policy.allocate(ids, [], 10)
+    assert len(out) == 10
+    dev_ids = [devices[i].dev_id for i in out]
+    assert len(set(dev_ids)) == 2
+
+
+def test_cpx_required_partition(fake_mi355x_cpx):
+    policy, devices = make_policy(fake_mi355x_cpx)
+    by_gpu = {}
+    for d in devices.values():
+        by_gpu.setdefault(d.dev_id, []).append(d.id)
+    target_gpu = sorted(by_gpu)[3]
+    required = [sorted(by_gpu[target_gpu])[0]]
+    out = policy.allocate(sorted(devices), required, 4)
+    assert required[0] in out
+    assert {devices[i].dev_id for i in out} == {target_gpu}
+
+
+def test_pair_weight_count_cpx(fake_mi355x_cpx):
+    """All-to-all mesh over 64 devices: 63 'from' keys (cf. reference
+    device_test.go:90-108 expecting n-1 on its mesh)."""
+    topo = KFDTopology.load(fake_mi355x_cpx.paths)
+    devices = list(discover_gpus(fake_mi355x_cpx.paths, topology=topo).values())
+    weights = compute_pair_weights(devices, topo)
+    assert len(weights) == 63
+    total_pairs = sum(len(v) for v in weights.values())
+    assert total_pairs == 64 * 63 // 2
+
+
+# ---------- hive awareness (MI355X-native extension) ----------
+
+def build_two_hive_node(root):
+    """8 GPUs, numa all 0, two xGMI hives of 4; cross-hive links also xGMI
+    so only the hive id discriminates."""
+    fs = FakeSysfs(root)
+    fs.add_cpu_node(0)
+    hives = [111, 222]
+    nodes = []
+    for i in range(8):
+        fs.add_physical_gpu(i, node_id=2 + i, numa_node=0, hive_id=hives[i // 4])
+        nodes.append(2 + i)
+    for a in range(8):
+        for b in range(a + 1, 8):
+            fs.add_link(nodes[a], nodes[b], link_type=11)
+    return fs
+
+
+def test_hive_packing(tmp_path):
+    fs = build_two_hive_node(str(tmp_path / "hive"))
+    policy, devices = make_policy(fs)
+    ids = sorted(devices)
+    # GPU0 (hive 1) excluded: naive order would pick {1,2,3,4} crossing
+    # hives; hive-aware weights must pick the intact hive {4,5,6,7}.
+    out = policy.allocate(ids[1:], [], 4)
+    hive2 = set(ids[4:])
+    assert set(out) == hive2, f"expected one-hive packing, got {out}"
+
+
+def test_init_requires_links(tmp_path):
+    fs = FakeSysfs(str(tmp_path / "nolinks"))
+    fs.add_cpu_node(0)
+    fs.add_physical_gpu(0, node_id=2)
+    fs.add_physical_gpu(1, node_id=3)
+    topo = KFDTopology.load(fs.paths)
+    devices = discover_gpus(fs.paths, topology=topo)
+    policy = BestEffortPolicy()
+    with pytest.raises(AllocationError):
+        policy.init(devices.values(), topology=topo)
+    with pytest.raises(AllocationError):
+        policy.init([], topology=topo)
