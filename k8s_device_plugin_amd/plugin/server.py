@@ -1,0 +1,208 @@
+"""DevicePlugin v1beta1 gRPC servicer.
+
+Parity with the reference's AMDGPUPlugin (reference:
+internal/pkg/plugin/plugin.go:41-397):
+  - ListAndWatch walks sysfs once, caches the device map for Allocate, sends
+    the initial list, then re-sends with refreshed health on every heartbeat;
+  - Allocate serves purely from the in-memory cache: /dev/kfd always plus
+    /dev/dri/card<N> + /dev/dri/renderD<M> per requested device — zero
+    syscalls on the pod-admission path;
+  - GetPreferredAllocation delegates to the hive-aware best-effort policy;
+  - heterogeneous nodes bucket devices by '<compute>_<memory>' and each
+    resource's plugin serves only its bucket.
+
+Deviation (improvement) from the reference: on ListAndWatch stream loss the
+reference os.Exit(1)s so the DaemonSet restarts the pod (plugin.go:322-324).
+We instead notify the lifecycle manager, which re-registers with the kubelet
+without killing the process; `exit_on_stream_loss=True` restores the
+reference behavior.
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import threading
+from typing import Callable, Dict, List, Optional
+
+from ..allocator import AllocationError, BestEffortPolicy
+from ..protos import deviceplugin as dp
+from ..protos import metricssvc as ms
+from ..topology import (
+    GPUDevice,
+    KFDTopology,
+    SysPaths,
+    discover_gpus,
+    is_homogeneous,
+    simple_health_check,
+)
+from ..health.exporter import populate_per_gpu_health
+
+log = logging.getLogger(__name__)
+
+
+class AMDGPUPlugin:
+    """One plugin instance per advertised resource (amd.com/<resource>)."""
+
+    def __init__(
+        self,
+        resource: str = "gpu",
+        paths: SysPaths = SysPaths(),
+        allocator: Optional[BestEffortPolicy] = None,
+        exporter_socket: str = ms.EXPORTER_SOCKET,
+        exporter_timeout: float = ms.QUERY_TIMEOUT_S,
+        on_stream_lost: Optional[Callable[[], None]] = None,
+        exit_on_stream_loss: bool = False,
+    ):
+        self.resource = resource
+        self.paths = paths
+        self.devices: Dict[str, GPUDevice] = {}
+        self.allocator = allocator or BestEffortPolicy()
+        self.allocator_init_error = False
+        self.exporter_socket = exporter_socket
+        self.exporter_timeout = exporter_timeout
+        self.on_stream_lost = on_stream_lost
+        self.exit_on_stream_loss = exit_on_stream_loss
+        self._cond = threading.Condition()
+        self._heartbeat_gen = 0
+        self._stop = threading.Event()
+
+    # ---- lifecycle ----
+
+    def start(self) -> None:
+        """Discover devices and init the allocator (reference: plugin.go:82-91).
+
+        Allocator failure degrades to kubelet-default allocation instead of
+        failing startup (plugin.go:86-89).
+        """
+        topo = KFDTopology.load(self.paths)
+        self.devices = discover_gpus(self.paths, topology=topo, strict=False)
+        try:
+            self.allocator.init(self.devices.values(), topology=topo)
+        except AllocationError as e:
+            log.error(
+                "allocator init failed, falling back to kubelet default "
+                "allocation: %s", e
+            )
+            self.allocator_init_error = True
+
+    def stop(self) -> None:
+        self._stop.set()
+        self.heartbeat()  # wake streams so they exit
+
+    def heartbeat(self) -> None:
+        """Trigger a health refresh on all ListAndWatch streams."""
+        with self._cond:
+            self._heartbeat_gen += 1
+            self._cond.notify_all()
+
+    # ---- DevicePlugin v1beta1 RPCs ----
+
+    def GetDevicePluginOptions(self, request, context):
+        opts = dp.DevicePluginOptions()
+        if not self.allocator_init_error:
+            opts.get_preferred_allocation_available = True
+        return opts
+
+    def PreStartContainer(self, request, context):
+        return dp.PreStartContainerResponse()
+
+    def _my_devices(self) -> List[GPUDevice]:
+        """Devices this resource's plugin advertises."""
+        if is_homogeneous(self.devices):
+            return list(self.devices.values())
+        return [
+            d for d in self.devices.values() if d.partition_key == self.resource
+        ]
+
+    def _device_list(self, health_default: Optional[str] = None) -> List:
+        out = []
+        for d in sorted(self._my_devices(), key=lambda x: x.id):
+            dev = dp.Device(ID=d.id, health=health_default or dp.HEALTHY)
+            dev.topology.nodes.add().ID = d.numa_node
+            out.append(dev)
+        return out
+
+    def ListAndWatch(self, request, context):
+        # re-walk sysfs and refresh the Allocate cache (plugin.go:231)
+        self.devices = discover_gpus(self.paths, strict=False)
+        log.info("found %d AMD GPU devices", len(self.devices))
+
+        devs = self._device_list()
+        yield dp.ListAndWatchResponse(devices=devs)
+
+        with self._cond:
+            gen = self._heartbeat_gen
+        while True:
+            with self._cond:
+                self._cond.wait_for(
+                    lambda: self._heartbeat_gen != gen or self._stop.is_set(),
+                    timeout=1.0,
+                )
+                fired = self._heartbeat_gen != gen
+                gen = self._heartbeat_gen
+            if self._stop.is_set():
+                return
+            if not context.is_active():
+                log.error(
+                    "ListAndWatch stream disconnected; triggering re-registration"
+                )
+                if self.exit_on_stream_loss:
+                    os._exit(1)
+                if self.on_stream_lost is not None:
+                    self.on_stream_lost()
+                return
+            if not fired:
+                continue
+
+            default = (
+                dp.HEALTHY if simple_health_check(self.paths) else dp.UNHEALTHY
+            )
+            devs = self._device_list(health_default=default)
+            populate_per_gpu_health(
+                devs, default, self.exporter_socket, self.exporter_timeout
+            )
+            yield dp.ListAndWatchResponse(devices=devs)
+
+    def GetPreferredAllocation(self, request, context):
+        import grpc
+
+        response = dp.PreferredAllocationResponse()
+        for req in request.container_requests:
+            try:
+                ids = self.allocator.allocate(
+                    list(req.available_deviceIDs),
+                    list(req.must_include_deviceIDs),
+                    int(req.allocation_size),
+                )
+            except AllocationError as e:
+                log.error("preferred allocation failed: %s", e)
+                context.abort(
+                    grpc.StatusCode.INVALID_ARGUMENT,
+                    f"unable to get preferred allocation list: {e}",
+                )
+                return response
+            response.container_responses.add().deviceIDs.extend(ids)
+        return response
+
+    def Allocate(self, request, context):
+        response = dp.AllocateResponse()
+        for req in request.container_requests:
+            car = response.container_responses.add()
+            # one /dev/kfd per node, always (plugin.go:368-374)
+            kfd = car.devices.add()
+            kfd.host_path = kfd.container_path = "/dev/kfd"
+            kfd.permissions = "rw"
+            for dev_id in req.devices_ids:
+                dev = self.devices.get(dev_id)
+                if dev is None:
+                    log.warning("Allocate: unknown device ID %s", dev_id)
+                    continue
+                for path in (
+                    f"/dev/dri/card{dev.card}",
+                    f"/dev/dri/renderD{dev.render_d}",
+                ):
+                    spec = car.devices.add()
+                    spec.host_path = spec.container_path = path
+                    spec.permissions = "rw"
+        return response
