@@ -1,0 +1,240 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: stub-kubelet pod admission on MI355X.
+
+Measures the BASELINE.json headline: Allocate() latency and
+advertised-vs-present GPU correctness at N GPUs requested.  One step = one
+pod admission for N GPUs: GetPreferredAllocation(size=N) + Allocate(N ids)
+against the live device plugin (real kfd sysfs on a GPU box; synthetic
+8*MI355X fake tree elsewhere).
+
+Contract (driver):
+  python bench.py --gpus N --steps K --warmup W
+prints ONE JSON line from rank 0; multi-rank runs come via torchrun with
+one rank per GPU (rank 0 drives the node-local plugin, every rank verifies
+its own GPU is present and participates in the timing barriers).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import tempfile
+import time
+
+
+def _dist_env():
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    return world, rank, local_rank
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=200)
+    ap.add_argument("--warmup", type=int, default=20)
+    ap.add_argument("--hbm-probe", action="store_true",
+                    help="also run the deep health probe before timing")
+    args = ap.parse_args()
+
+    import torch
+
+    world, rank, local_rank = _dist_env()
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        dist.init_process_group(backend="gloo")
+
+    cuda = torch.cuda.is_available()
+
+    # ---- advertised-vs-present: every rank touches its own GPU ----
+    present_gpus = torch.cuda.device_count() if cuda else 0
+    if cuda:
+        torch.cuda.set_device(local_rank)
+        x = torch.ones(1024, 1024, device=f"cuda:{local_rank}")
+        y = (x @ x).sum()
+        assert float(y) == 1024.0 * 1024 * 1024, "GPU sanity matmul failed"
+        torch.cuda.synchronize()
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        if cuda:
+            torch.cuda.synchronize()
+
+    result = {}
+    harness = None
+    if rank == 0:
+        harness = _Harness(args.gpus)
+        harness.start()
+        # warmup
+        for _ in range(args.warmup):
+            harness.step()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    if rank == 0:
+        for _ in range(args.steps):
+            harness.step()
+    barrier_sync()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t[0])
+
+    if rank == 0:
+        alloc_p50_us = statistics.median(harness.alloc_lat_us)
+        pref_p50_us = (
+            statistics.median(harness.pref_lat_us) if harness.pref_lat_us else None
+        )
+        advertised = harness.advertised
+        harness.stop()
+
+        value = args.gpus * args.steps / elapsed  # device-grants/s, whole job
+        out = {
+            "metric": "pod_admission_device_grants_per_s",
+            "value": round(value, 2),
+            "unit": "device-grants/s",
+            "n_gpus": args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 4),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "n/a",
+            "data": "synthetic",
+            "config": {
+                "model": "stub-kubelet-pod-admission",
+                "resource": "amd.com/gpu",
+                "devices_per_allocate": args.gpus,
+                "parallelism": "node-local",
+                "sysfs": harness.sysfs_kind,
+                "advertised_gpus": advertised,
+                "present_gpus": present_gpus,
+                "allocate_p50_us": round(alloc_p50_us, 1),
+                "preferred_alloc_p50_us": (
+                    round(pref_p50_us, 1) if pref_p50_us is not None else None
+                ),
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+    if dist is not None:
+        dist.barrier()
+        dist.destroy_process_group()
+    return 0
+
+
+class _Harness:
+    """Plugin + stub kubelet over live or synthetic sysfs (rank 0 only)."""
+
+    def __init__(self, n_gpus: int):
+        self.n = n_gpus
+        self.alloc_lat_us = []
+        self.pref_lat_us = []
+        self.advertised = 0
+        self.sysfs_kind = "unknown"
+        self._tmp = None
+        self._mgr = None
+        self._kubelet = None
+        self._stream_call = None
+
+    def start(self) -> None:
+        from k8s_device_plugin_amd.plugin import AMDGPUPlugin, PluginManager
+        from k8s_device_plugin_amd.protos import deviceplugin as dp
+        from k8s_device_plugin_amd.testing.stub_kubelet import StubKubelet
+        from k8s_device_plugin_amd.topology import (
+            SysPaths,
+            discover_gpus,
+            simple_health_check,
+        )
+
+        self._tmp = tempfile.TemporaryDirectory(prefix="amdxdp-bench-")
+        root = self._tmp.name
+
+        live = SysPaths("/")
+        if os.path.isdir(live.kfd_class) and simple_health_check(live):
+            paths = live
+            self.sysfs_kind = "live"
+        else:
+            from k8s_device_plugin_amd.testing.fakesysfs import build_mi355x_node
+
+            fs = build_mi355x_node(
+                os.path.join(root, "fakesys"), n_gpus=max(self.n, 8)
+            )
+            paths = fs.paths
+            self.sysfs_kind = "fake-8xMI355X"
+
+        dp_dir = os.path.join(root, "device-plugins")
+        self._kubelet = StubKubelet(dp_dir).start()
+        self._mgr = PluginManager(
+            lambda res: AMDGPUPlugin(resource=res, paths=paths),
+            device_plugin_path=dp_dir,
+        )
+        self._mgr.run(["gpu"])
+        reg = self._kubelet.wait_for_registration()
+        self._stub = self._kubelet.connect(reg.endpoint)
+        self._dp = dp
+
+        # initial ListAndWatch: the advertised device set
+        call = self._stub.ListAndWatch(dp.Empty())
+        first = next(iter(call))
+        self._stream_call = call
+        self.device_ids = sorted(d.ID for d in first.devices)
+        self.advertised = len(self.device_ids)
+        if self.advertised < self.n:
+            raise RuntimeError(
+                f"advertised {self.advertised} devices < requested {self.n}"
+            )
+        self._preferred_available = not any(
+            d.health != "Healthy" for d in first.devices
+        )
+
+    def step(self) -> None:
+        dp = self._dp
+        # 1. GetPreferredAllocation for N devices
+        req = dp.PreferredAllocationRequest()
+        cr = req.container_requests.add()
+        cr.available_deviceIDs.extend(self.device_ids)
+        cr.allocation_size = self.n
+        t0 = time.perf_counter()
+        resp = self._stub.GetPreferredAllocation(req, timeout=10)
+        t1 = time.perf_counter()
+        self.pref_lat_us.append((t1 - t0) * 1e6)
+        chosen = list(resp.container_responses[0].deviceIDs)
+
+        # 2. Allocate them
+        areq = dp.AllocateRequest()
+        areq.container_requests.add().devices_ids.extend(chosen)
+        t0 = time.perf_counter()
+        aresp = self._stub.Allocate(areq, timeout=10)
+        t1 = time.perf_counter()
+        self.alloc_lat_us.append((t1 - t0) * 1e6)
+        specs = aresp.container_responses[0].devices
+        assert len(specs) == 1 + 2 * self.n, (
+            f"expected /dev/kfd + 2 nodes per device, got {len(specs)}"
+        )
+
+    def stop(self) -> None:
+        if self._stream_call is not None:
+            self._stream_call.cancel()
+        if self._mgr is not None:
+            self._mgr.stop()
+        if self._kubelet is not None:
+            self._kubelet.stop()
+        if self._tmp is not None:
+            self._tmp.cleanup()
+
+
+if __name__ == "__main__":
+    sys.exit(main())

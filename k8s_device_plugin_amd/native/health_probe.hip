@@ -1,0 +1,259 @@
+// gfx950 deep health probe (HIP/CDNA4).
+//
+// MI355X-native upgrade over the reference's liveness check, which only
+// opens the device node (reference: internal/pkg/amdgpu/amdgpu.go:390-399).
+// This probe actually executes CDNA4 code on the GPU and verifies:
+//   1. wavefront size is 64 and all lanes participate (ballot);
+//   2. the bf16 MFMA matrix pipe computes correct sums
+//      (v_mfma_f32_16x16x32_bf16 invariant checks, exact in bf16/f32);
+//   3. LDS write/read round-trips a full 128 KiB per CU;
+//   4. HBM sustains a float4 streaming copy at a reportable GB/s
+//      (grid sized >>256 workgroups to cover all 8 XCDs).
+//
+// Used by the plugin's optional deep health check, __graft_entry__.smoke(),
+// and the gpu-marked tests.  Build: hipcc --offload-arch=gfx950, see
+// native/build.py.  Guide: /opt/skills/guides/cdna_hip_programming.md §3.
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+namespace py = pybind11;
+
+#define HIP_CHECK(expr)                                                        \
+    do {                                                                       \
+        hipError_t _e = (expr);                                                \
+        if (_e != hipSuccess)                                                  \
+            throw std::runtime_error(std::string("HIP error at " #expr ": ") + \
+                                     hipGetErrorString(_e));                   \
+    } while (0)
+
+namespace {
+
+// ---------------- wavefront probe ----------------
+
+__global__ void wave_probe_kernel(unsigned long long *out) {
+    unsigned long long active = __ballot(1);
+    if (threadIdx.x == 0) {
+        out[0] = active;
+        out[1] = warpSize;
+    }
+}
+
+// ---------------- MFMA probe ----------------
+
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+__device__ inline __bf16 small_int_bf16(int v) {
+    return (__bf16)(float)v;  // small ints are exact in bf16
+}
+
+// One wave computes D = A*B with v_mfma_f32_16x16x32_bf16 twice:
+//  pass 0: A == 1, B == 0.5     -> every D element must equal 16.0
+//  pass 1: A[lane][r] = pattern, B == 1 -> sum(D) == 16 * sum(A) (exact)
+__global__ void mfma_probe_kernel(float *d_out /* [2][256] */) {
+#if defined(__gfx950__)
+    int lane = threadIdx.x;
+    if (lane >= 64) return;
+
+    for (int pass = 0; pass < 2; ++pass) {
+        bf16x8 a, b;
+        for (int r = 0; r < 8; ++r) {
+            if (pass == 0) {
+                a[r] = small_int_bf16(1);
+                b[r] = (__bf16)0.5f;
+            } else {
+                a[r] = small_int_bf16(((lane * 8 + r) % 7) - 3);
+                b[r] = small_int_bf16(1);
+            }
+        }
+        f32x4 c = {0.f, 0.f, 0.f, 0.f};
+        f32x4 d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+        for (int r = 0; r < 4; ++r)
+            d_out[pass * 256 + lane * 4 + r] = d[r];
+    }
+#else
+    if (threadIdx.x == 0) d_out[0] = -1.0f;
+#endif
+}
+
+// ---------------- LDS probe ----------------
+
+__global__ void lds_probe_kernel(uint32_t *err_count, int lds_words) {
+    extern __shared__ uint32_t lds[];
+    int tid = threadIdx.x;
+    int nthreads = blockDim.x;
+    for (int i = tid; i < lds_words; i += nthreads)
+        lds[i] = (uint32_t)i * 2654435761u + blockIdx.x;
+    __syncthreads();
+    // read back with a stride so each thread checks other threads' writes
+    uint32_t errors = 0;
+    for (int i = tid * 17 % lds_words, n = 0; n < lds_words; ++n, i = (i + 1) % lds_words)
+        if (lds[i] != (uint32_t)i * 2654435761u + blockIdx.x) ++errors;
+    if (errors) atomicAdd(err_count, errors);
+}
+
+// ---------------- HBM streaming copy ----------------
+
+__global__ void hbm_copy_kernel(const float4 *__restrict__ src,
+                                float4 *__restrict__ dst, size_t n) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) dst[i] = src[i];
+}
+
+__global__ void fill_pattern_kernel(float4 *buf, size_t n) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        float v = (float)(i & 0xFFFF);
+        buf[i] = make_float4(v, v + 0.25f, v + 0.5f, v + 0.75f);
+    }
+}
+
+// ---------------- host-side probe ----------------
+
+py::dict run_probe(int device, size_t hbm_bytes) {
+    py::dict result;
+    HIP_CHECK(hipSetDevice(device));
+
+    hipDeviceProp_t props;
+    HIP_CHECK(hipGetDeviceProperties(&props, device));
+    result["device_name"] = std::string(props.name);
+    result["gcn_arch"] = std::string(props.gcnArchName);
+    result["cu_count"] = props.multiProcessorCount;
+    result["vram_bytes"] = (uint64_t)props.totalGlobalMem;
+
+    // 1. wavefront
+    {
+        unsigned long long *d;
+        HIP_CHECK(hipMalloc(&d, 2 * sizeof(unsigned long long)));
+        hipLaunchKernelGGL(wave_probe_kernel, dim3(1), dim3(64), 0, 0, d);
+        HIP_CHECK(hipGetLastError());
+        unsigned long long h[2];
+        HIP_CHECK(hipMemcpy(h, d, sizeof(h), hipMemcpyDeviceToHost));
+        HIP_CHECK(hipFree(d));
+        result["wavefront_size"] = (int)h[1];
+        result["wave_ok"] = (h[0] == ~0ull && h[1] == 64);
+    }
+
+    // 2. MFMA
+    {
+        float *d;
+        HIP_CHECK(hipMalloc(&d, 2 * 256 * sizeof(float)));
+        HIP_CHECK(hipMemset(d, 0, 2 * 256 * sizeof(float)));
+        hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, 0, d);
+        HIP_CHECK(hipGetLastError());
+        std::vector<float> h(2 * 256);
+        HIP_CHECK(hipMemcpy(h.data(), d, h.size() * sizeof(float),
+                            hipMemcpyDeviceToHost));
+        HIP_CHECK(hipFree(d));
+
+        bool pass0 = true;
+        for (int i = 0; i < 256; ++i)
+            if (h[i] != 16.0f) pass0 = false;
+        double sum_a = 0;
+        for (int lane = 0; lane < 64; ++lane)
+            for (int r = 0; r < 8; ++r)
+                sum_a += ((lane * 8 + r) % 7) - 3;
+        double sum_d = 0;
+        for (int i = 0; i < 256; ++i) sum_d += h[256 + i];
+        bool pass1 = (sum_d == 16.0 * sum_a);
+        result["mfma_ok"] = pass0 && pass1;
+        result["mfma_const_ok"] = pass0;
+        result["mfma_sum_ok"] = pass1;
+    }
+
+    // 3. LDS: 128 KiB dynamic per workgroup (160 KiB/CU on gfx950)
+    {
+        const int lds_bytes = 128 * 1024;
+        uint32_t *d_err;
+        HIP_CHECK(hipMalloc(&d_err, sizeof(uint32_t)));
+        HIP_CHECK(hipMemset(d_err, 0, sizeof(uint32_t)));
+        hipLaunchKernelGGL(lds_probe_kernel, dim3(props.multiProcessorCount),
+                           dim3(256), lds_bytes, 0, d_err, lds_bytes / 4);
+        HIP_CHECK(hipGetLastError());
+        uint32_t h_err = 1;
+        HIP_CHECK(hipMemcpy(&h_err, d_err, sizeof(h_err), hipMemcpyDeviceToHost));
+        HIP_CHECK(hipFree(d_err));
+        result["lds_ok"] = (h_err == 0);
+        result["lds_bytes_tested"] = lds_bytes;
+    }
+
+    // 4. HBM streaming bandwidth (read+write)
+    {
+        size_t n = hbm_bytes / sizeof(float4);
+        float4 *src, *dst;
+        HIP_CHECK(hipMalloc(&src, n * sizeof(float4)));
+        HIP_CHECK(hipMalloc(&dst, n * sizeof(float4)));
+        // >>256 workgroups to fill all 8 XCDs (256 CUs)
+        const int blocks = 8192, threads = 256, iters = 5;
+        hipLaunchKernelGGL(fill_pattern_kernel, dim3(blocks), dim3(threads), 0, 0,
+                           src, n);
+        hipLaunchKernelGGL(hbm_copy_kernel, dim3(blocks), dim3(threads), 0, 0,
+                           src, dst, n);  // warmup
+        HIP_CHECK(hipDeviceSynchronize());
+
+        hipEvent_t t0, t1;
+        HIP_CHECK(hipEventCreate(&t0));
+        HIP_CHECK(hipEventCreate(&t1));
+        HIP_CHECK(hipEventRecord(t0));
+        for (int i = 0; i < iters; ++i)
+            hipLaunchKernelGGL(hbm_copy_kernel, dim3(blocks), dim3(threads), 0, 0,
+                               src, dst, n);
+        HIP_CHECK(hipEventRecord(t1));
+        HIP_CHECK(hipEventSynchronize(t1));
+        float ms = 0;
+        HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+        double gbps = (2.0 * n * sizeof(float4) * iters) / (ms * 1e6);
+        result["hbm_gbps"] = gbps;
+        result["hbm_bytes_tested"] = (uint64_t)(n * sizeof(float4));
+
+        // spot-check correctness of the copy
+        std::vector<float4> sample(16);
+        HIP_CHECK(hipMemcpy(sample.data(), dst + n / 2,
+                            sample.size() * sizeof(float4),
+                            hipMemcpyDeviceToHost));
+        bool copy_ok = true;
+        for (size_t k = 0; k < sample.size(); ++k) {
+            size_t i = n / 2 + k;
+            if (sample[k].x != (float)(i & 0xFFFF)) copy_ok = false;
+        }
+        result["hbm_copy_ok"] = copy_ok;
+        HIP_CHECK(hipEventDestroy(t0));
+        HIP_CHECK(hipEventDestroy(t1));
+        HIP_CHECK(hipFree(src));
+        HIP_CHECK(hipFree(dst));
+    }
+
+    bool healthy = result["wave_ok"].cast<bool>() &&
+                   result["mfma_ok"].cast<bool>() &&
+                   result["lds_ok"].cast<bool>() &&
+                   result["hbm_copy_ok"].cast<bool>();
+    result["healthy"] = healthy;
+    return result;
+}
+
+int device_count() {
+    int n = 0;
+    hipError_t e = hipGetDeviceCount(&n);
+    if (e != hipSuccess) return 0;
+    return n;
+}
+
+} // namespace
+
+PYBIND11_MODULE(_healthprobe, m) {
+    m.doc() = "gfx950 deep GPU health probe (MFMA/LDS/HBM)";
+    m.def("run_probe", &run_probe, py::arg("device") = 0,
+          py::arg("hbm_bytes") = (size_t)1 << 30);
+    m.def("device_count", &device_count);
+}
