@@ -185,33 +185,41 @@ class _Harness:
         reg = self._kubelet.wait_for_registration()
         self._stub = self._kubelet.connect(reg.endpoint)
         self._dp = dp
+        # on a 1-GPU node there are no GPU-GPU links, so the plugin degrades
+        # to kubelet-default allocation (no GetPreferredAllocation) exactly
+        # like the reference (plugin.go:86-89,210-217)
+        self.preferred_available = reg.options.get_preferred_allocation_available
 
         # initial ListAndWatch: the advertised device set
         call = self._stub.ListAndWatch(dp.Empty())
         first = next(iter(call))
         self._stream_call = call
-        self.device_ids = sorted(d.ID for d in first.devices)
+        # only Healthy (kfd-backed) devices are schedulable; cgroup-masked
+        # peer GPUs are advertised Unhealthy on restricted boxes
+        self.device_ids = sorted(
+            d.ID for d in first.devices if d.health == "Healthy"
+        )
         self.advertised = len(self.device_ids)
         if self.advertised < self.n:
             raise RuntimeError(
-                f"advertised {self.advertised} devices < requested {self.n}"
+                f"advertised {self.advertised} healthy devices < requested {self.n}"
             )
-        self._preferred_available = not any(
-            d.health != "Healthy" for d in first.devices
-        )
 
     def step(self) -> None:
         dp = self._dp
-        # 1. GetPreferredAllocation for N devices
-        req = dp.PreferredAllocationRequest()
-        cr = req.container_requests.add()
-        cr.available_deviceIDs.extend(self.device_ids)
-        cr.allocation_size = self.n
-        t0 = time.perf_counter()
-        resp = self._stub.GetPreferredAllocation(req, timeout=10)
-        t1 = time.perf_counter()
-        self.pref_lat_us.append((t1 - t0) * 1e6)
-        chosen = list(resp.container_responses[0].deviceIDs)
+        # 1. GetPreferredAllocation for N devices (when advertised)
+        if self.preferred_available:
+            req = dp.PreferredAllocationRequest()
+            cr = req.container_requests.add()
+            cr.available_deviceIDs.extend(self.device_ids)
+            cr.allocation_size = self.n
+            t0 = time.perf_counter()
+            resp = self._stub.GetPreferredAllocation(req, timeout=10)
+            t1 = time.perf_counter()
+            self.pref_lat_us.append((t1 - t0) * 1e6)
+            chosen = list(resp.container_responses[0].deviceIDs)
+        else:
+            chosen = self.device_ids[: self.n]
 
         # 2. Allocate them
         areq = dp.AllocateRequest()

@@ -76,7 +76,14 @@ def compute_pair_weights(
     device.go:160-253).  Missing pairs score 0 when summed, matching the
     reference's map-default semantics.
     """
-    by_node: Dict[int, GPUDevice] = {d.node_id: d for d in devices}
+    # only devices whose node_id genuinely maps to their kfd topology node
+    # participate (guards against placeholder node_id values from devices
+    # the container's cgroup masks out of the topology)
+    by_node: Dict[int, GPUDevice] = {}
+    for d in devices:
+        node = topology.nodes.get(d.node_id)
+        if node is not None and node.render_minor == d.render_d:
+            by_node[d.node_id] = d
     weights: Dict[int, Dict[int, int]] = {}
     for node in topology.nodes.values():
         if node.render_minor <= 0:

@@ -77,8 +77,9 @@ class AMDGPUPlugin:
         """
         topo = KFDTopology.load(self.paths)
         self.devices = discover_gpus(self.paths, topology=topo, strict=False)
+        schedulable = [d for d in self.devices.values() if d.kfd_backed]
         try:
-            self.allocator.init(self.devices.values(), topology=topo)
+            self.allocator.init(schedulable, topology=topo)
         except AllocationError as e:
             log.error(
                 "allocator init failed, falling back to kubelet default "
@@ -118,7 +119,10 @@ class AMDGPUPlugin:
     def _device_list(self, health_default: Optional[str] = None) -> List:
         out = []
         for d in sorted(self._my_devices(), key=lambda x: x.id):
-            dev = dp.Device(ID=d.id, health=health_default or dp.HEALTHY)
+            # a PCI-visible GPU with no kfd topology node (cgroup-masked
+            # peer) is not schedulable: advertise it Unhealthy
+            health = (health_default or dp.HEALTHY) if d.kfd_backed else dp.UNHEALTHY
+            dev = dp.Device(ID=d.id, health=health)
             dev.topology.nodes.add().ID = d.numa_node
             out.append(dev)
         return out
@@ -129,10 +133,12 @@ class AMDGPUPlugin:
         log.info("found %d AMD GPU devices", len(self.devices))
 
         devs = self._device_list()
-        yield dp.ListAndWatchResponse(devices=devs)
-
+        # snapshot the heartbeat generation BEFORE the first send: a beat
+        # firing while the initial response is in flight must not be lost
         with self._cond:
             gen = self._heartbeat_gen
+        yield dp.ListAndWatchResponse(devices=devs)
+
         while True:
             with self._cond:
                 self._cond.wait_for(
@@ -162,6 +168,12 @@ class AMDGPUPlugin:
             populate_per_gpu_health(
                 devs, default, self.exporter_socket, self.exporter_timeout
             )
+            # an exporter verdict cannot resurrect a device the kfd
+            # topology does not back
+            unbacked = {d.id for d in self._my_devices() if not d.kfd_backed}
+            for dev in devs:
+                if dev.ID in unbacked:
+                    dev.health = dp.UNHEALTHY
             yield dp.ListAndWatchResponse(devices=devs)
 
     def GetPreferredAllocation(self, request, context):

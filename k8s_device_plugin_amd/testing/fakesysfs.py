@@ -16,7 +16,7 @@ from typing import Dict, List, Optional
 from ..topology.sysfs import SysPaths
 
 # MI355X (gfx950) defaults
-MI355X_VRAM_BYTES = 288 * 1024**3          # 288 GB HBM3E
+MI355X_VRAM_BYTES = 309220868096           # 288 GB HBM3E (value read from a real MI355X kfd node)
 MI355X_SIMD_COUNT = 1024                   # 256 CUs x 4 SIMDs
 MI355X_SIMD_PER_CU = 4
 MI355X_GFX_TARGET = 90500
@@ -131,7 +131,7 @@ class FakeSysfs:
         gfx_target_version: int = MI355X_GFX_TARGET,
         vram_bytes: int = MI355X_VRAM_BYTES,
         numa_node: int = 0,
-        device_id: int = 0x75A0,
+        device_id: int = 0x75A3,
     ) -> None:
         self.add_kfd_node(
             node_id,
@@ -181,8 +181,8 @@ class FakeSysfs:
         numa_node: int = 0,
         hive_id: int = 0,
         partition_caps: bool = True,
-        product: str = "AMD Instinct MI355X",
-        device_id: int = 0x75A0,
+        product: str = "AMD Instinct MI355 OAM",
+        device_id: int = 0x75A3,
         **gpu_node_kwargs,
     ) -> str:
         """Add one physical GPU: PCI dir + drm minors + its kfd node.

@@ -41,6 +41,12 @@ class GPUDevice:
     memory_partition: str = ""    # e.g. "nps1", "nps4" (lowercased)
     numa_node: int = -1
     node_id: int = 0         # kfd topology node index
+    # False when the device's renderD has no readable kfd topology node —
+    # e.g. a peer GPU masked away by the container's cgroup (observed on
+    # MI355X boxes: peer nodes' properties read "Operation not permitted").
+    # Such a device is visible on PCI but not schedulable; the plugin
+    # advertises it Unhealthy.
+    kfd_backed: bool = True
 
     @property
     def is_partition(self) -> bool:
@@ -127,9 +133,11 @@ def discover_gpus(
             memory_partition=(memory or "").lower(),
             numa_node=numa_node,
         )
-        if render_d is not None:
-            dev.dev_id = rd_to_devid.get(render_d, "")
+        if render_d is not None and render_d in rd_to_devid:
+            dev.dev_id = rd_to_devid[render_d]
             dev.node_id = rd_to_nodeid.get(render_d, 0)
+        else:
+            dev.kfd_backed = False
         devices[pci_addr] = dev
 
     # --- compute partitions (amdgpu_xcp platform devices) ---

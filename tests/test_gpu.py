@@ -48,7 +48,11 @@ def test_live_kfd_discovery(live_devices):
         if node.properties.get("num_xcc", 0) >= 8:
             assert node.vram_bytes > 280 * 1024**3
             assert node.cu_count == 256
-    for d in devs.values():
+    # gpurun containers cgroup-mask peer GPUs' kfd nodes: only kfd-backed
+    # devices carry the devID join
+    backed = [d for d in devs.values() if d.kfd_backed]
+    assert backed, "no kfd-backed device on a GPU box"
+    for d in backed:
         assert d.render_d >= 128
         assert d.dev_id, f"devID join failed for {d.id}"
 
@@ -58,10 +62,11 @@ def test_advertised_vs_present(live_devices):
 
     _, devs = live_devices
     present = torch.cuda.device_count()
-    # partitions can multiply advertised devices; physical count must match
-    physical = len({d.dev_id for d in devs.values()})
+    # schedulable (kfd-backed) physical GPUs must match what HIP sees;
+    # cgroup-masked peers are advertised Unhealthy and don't count
+    physical = len({d.dev_id for d in devs.values() if d.kfd_backed})
     assert physical == present, (
-        f"kfd walk found {physical} physical GPUs, torch sees {present}"
+        f"kfd walk found {physical} schedulable GPUs, torch sees {present}"
     )
 
 
@@ -70,7 +75,8 @@ def test_drmctl_ioctls(live_devices):
 
     _, devs = live_devices
     drm = load_drmctl(required=True)
-    d = sorted(devs.values(), key=lambda x: x.render_d)[0]
+    backed = [d for d in devs.values() if d.kfd_backed]
+    d = sorted(backed, key=lambda x: x.render_d)[0]
     dev_path = f"/dev/dri/renderD{d.render_d}"
 
     assert drm.dev_functional(dev_path)

@@ -1,0 +1,67 @@
+"""Restricted-visibility containers: PCI-visible GPUs whose kfd nodes are
+masked (observed on shared MI355X boxes: peer nodes' properties are
+unreadable).  Such devices must be advertised Unhealthy and excluded from
+the allocator."""
+
+import os
+import shutil
+
+from k8s_device_plugin_amd.plugin import AMDGPUPlugin
+from k8s_device_plugin_amd.protos import deviceplugin as dp
+from k8s_device_plugin_amd.topology import discover_gpus
+from k8s_device_plugin_amd.testing.fakesysfs import build_mi355x_node
+
+
+class _Ctx:
+    def is_active(self):
+        return True
+
+
+def _mask_kfd_nodes(fs, keep_node_ids):
+    nodes_dir = fs.paths.kfd_topology_nodes
+    for name in os.listdir(nodes_dir):
+        if int(name) not in keep_node_ids:
+            shutil.rmtree(os.path.join(nodes_dir, name))
+
+
+def test_masked_peers_unhealthy(tmp_path):
+    fs = build_mi355x_node(str(tmp_path / "r"), n_gpus=8)
+    # container sees CPUs (0,1) and only GPU node 2; peers masked
+    _mask_kfd_nodes(fs, {0, 1, 2})
+
+    devs = discover_gpus(fs.paths)
+    assert len(devs) == 8
+    backed = [d for d in devs.values() if d.kfd_backed]
+    assert len(backed) == 1
+    assert backed[0].dev_id
+
+    plugin = AMDGPUPlugin(resource="gpu", paths=fs.paths)
+    plugin.start()
+    # 1 visible GPU -> no GPU-GPU links -> allocator degrades like the
+    # reference on a 1-GPU node
+    assert plugin.allocator_init_error
+    opts = plugin.GetDevicePluginOptions(dp.Empty(), None)
+    assert not opts.get_preferred_allocation_available
+
+    stream = plugin.ListAndWatch(dp.Empty(), _Ctx())
+    first = next(stream)
+    health = {d.ID: d.health for d in first.devices}
+    assert sum(1 for h in health.values() if h == "Healthy") == 1
+    assert sum(1 for h in health.values() if h == "Unhealthy") == 7
+    plugin.stop()
+
+
+def test_masked_peers_heartbeat_keeps_unbacked_unhealthy(tmp_path):
+    fs = build_mi355x_node(str(tmp_path / "r"), n_gpus=4)
+    _mask_kfd_nodes(fs, {0, 1, 2, 3})  # two GPUs visible
+
+    plugin = AMDGPUPlugin(resource="gpu", paths=fs.paths)
+    plugin.start()
+    stream = plugin.ListAndWatch(dp.Empty(), _Ctx())
+    next(stream)
+    plugin.heartbeat()
+    resp = next(stream)
+    health = {d.ID: d.health for d in resp.devices}
+    assert sum(1 for h in health.values() if h == "Healthy") == 2
+    assert sum(1 for h in health.values() if h == "Unhealthy") == 2
+    plugin.stop()
