@@ -1,0 +1,169 @@
+"""CLI entry points: device plugin and node labeller daemons.
+
+Flag surface matches the reference binaries:
+  device plugin: -pulse, -resource_naming_strategy
+    (reference: cmd/k8s-device-plugin/main.go:107-112)
+  labeller: one boolean flag per label kind + DS_NODE_NAME env
+    (reference: cmd/k8s-node-labeller/main.go:407-409,440)
+Exit codes: 2 when the amdgpu/kfd driver is unavailable (reference:
+amdgpu.go:157-160).
+"""
+
+from __future__ import annotations
+
+import argparse
+import logging
+import os
+import signal
+import sys
+import threading
+import time
+
+from . import __version__
+from .topology import SysPaths, discover_gpus, DriverUnavailableError
+
+
+def _setup_logging(verbose: int) -> None:
+    logging.basicConfig(
+        level=logging.DEBUG if verbose else logging.INFO,
+        format="%(asctime)s %(levelname).1s %(name)s: %(message)s",
+    )
+
+
+def device_plugin_main(argv=None) -> int:
+    ap = argparse.ArgumentParser(
+        prog="amd-device-plugin",
+        description=f"AMD GPU device plugin for Kubernetes (MI355X-native) v{__version__}",
+    )
+    ap.add_argument("-pulse", "--pulse", type=int, default=0,
+                    help="seconds between health checks (0 disables)")
+    ap.add_argument("-resource_naming_strategy", "--resource_naming_strategy",
+                    default="single", help="single or mixed")
+    ap.add_argument("--kubelet-dir", default=None,
+                    help="device-plugin dir override (tests)")
+    ap.add_argument("--sysroot", default="/", help="sysfs root override (tests)")
+    ap.add_argument("-v", "--verbose", action="count", default=0)
+    args = ap.parse_args(argv)
+    _setup_logging(args.verbose)
+    log = logging.getLogger("amd-device-plugin")
+
+    from .health import HeartbeatTicker
+    from .plugin import (
+        AMDGPUPlugin,
+        PluginManager,
+        StrategyError,
+        get_resource_list,
+        parse_strategy,
+    )
+    from .protos import deviceplugin as dp
+
+    try:
+        strategy = parse_strategy(args.resource_naming_strategy)
+    except StrategyError as e:
+        log.error("%s", e)
+        return 1
+
+    paths = SysPaths(args.sysroot)
+
+    # gate on the ROCm driver being present (reference: main.go:139-152)
+    deadline = time.monotonic() + 60
+    while not os.path.isdir(paths.kfd_class):
+        if time.monotonic() > deadline:
+            log.error("/sys/class/kfd not present; is the amdgpu driver loaded? (exit 2)")
+            return 2
+        log.warning("waiting for %s ...", paths.kfd_class)
+        time.sleep(2)
+
+    try:
+        devices = discover_gpus(paths)
+    except DriverUnavailableError as e:
+        log.error("%s (exit 2)", e)
+        return 2
+
+    try:
+        resources = get_resource_list(devices, strategy)
+    except StrategyError as e:
+        log.error("%s", e)
+        return 1
+    if not resources:
+        log.error("no AMD GPUs found; nothing to advertise")
+        return 0
+    log.info("advertising resources: %s", [f"amd.com/{r}" for r in resources])
+
+    mgr = PluginManager(
+        lambda res: AMDGPUPlugin(resource=res, paths=paths),
+        device_plugin_path=args.kubelet_dir or dp.DEVICE_PLUGIN_PATH,
+    )
+    ticker = HeartbeatTicker(args.pulse)
+    ticker.subscribe(mgr.heartbeat_all)
+
+    stop = threading.Event()
+    for sig in (signal.SIGINT, signal.SIGTERM, signal.SIGQUIT):
+        signal.signal(sig, lambda *a: stop.set())
+
+    mgr.run(resources)
+    ticker.start()
+    try:
+        while not stop.wait(1.0):
+            pass
+    finally:
+        log.info("shutting down")
+        ticker.stop()
+        mgr.stop()
+    return 0
+
+
+def labeller_main(argv=None) -> int:
+    from .labeller import K8sClient, NodeLabelController, generate_labels
+    from .labeller.labels import LABEL_KINDS
+
+    ap = argparse.ArgumentParser(
+        prog="amd-node-labeller",
+        description=f"AMD GPU node labeller for Kubernetes (MI355X-native) v{__version__}",
+    )
+    for kind in LABEL_KINDS:
+        flag = kind.replace("-", "_")
+        ap.add_argument(f"-{flag}", f"--{flag}", action="store_true",
+                        help=f"label nodes with {kind} properties")
+    ap.add_argument("--sysroot", default="/", help="sysfs root override (tests)")
+    ap.add_argument("--api-server", default=None,
+                    help="k8s API base URL override (tests)")
+    ap.add_argument("--oneshot", action="store_true",
+                    help="reconcile once and exit (no watch)")
+    ap.add_argument("-v", "--verbose", action="count", default=0)
+    args = ap.parse_args(argv)
+    _setup_logging(args.verbose)
+    log = logging.getLogger("amd-node-labeller")
+
+    node_name = os.environ.get("DS_NODE_NAME") or os.environ.get("NODE_NAME")
+    if not node_name:
+        log.error("DS_NODE_NAME env var not set")
+        return 1
+
+    enabled = {kind: getattr(args, kind.replace("-", "_")) for kind in LABEL_KINDS}
+    if not any(enabled.values()):
+        log.warning("no label kinds enabled; pass e.g. --vram --family --cu_count")
+
+    paths = SysPaths(args.sysroot)
+    labels = generate_labels(enabled, paths)
+    log.info("computed %d label(s)", len(labels))
+
+    client = K8sClient(base_url=args.api_server)
+    ctl = NodeLabelController(client, node_name, labels)
+
+    stop = threading.Event()
+    for sig in (signal.SIGINT, signal.SIGTERM):
+        signal.signal(sig, lambda *a: (stop.set(), ctl.stop()))
+
+    if args.oneshot:
+        ctl.reconcile()
+        return 0
+    ctl.run(block=True)
+    return 0
+
+
+if __name__ == "__main__":
+    prog = os.path.basename(sys.argv[0])
+    if "labeller" in prog or (len(sys.argv) > 1 and sys.argv[1] == "labeller"):
+        sys.exit(labeller_main(sys.argv[2:] if sys.argv[1:2] == ["labeller"] else None))
+    sys.exit(device_plugin_main())

@@ -1,0 +1,160 @@
+"""Node labeller tests: frozen label-key inventory, value computation on the
+fake MI355X tree, cleanup semantics, and end-to-end reconcile against a
+fake API server (reference test model: cmd/k8s-node-labeller/main_test.go)."""
+
+import pytest
+
+from k8s_device_plugin_amd.labeller import (
+    LABEL_KINDS,
+    create_label_prefix,
+    generate_labels,
+    remove_old_node_labels,
+)
+from k8s_device_plugin_amd.labeller.k8s import K8sClient
+from k8s_device_plugin_amd.labeller.controller import NodeLabelController
+from k8s_device_plugin_amd.labeller.labels import product_name_from_ids
+from k8s_device_plugin_amd.testing.fake_k8s import FakeK8s
+
+
+def test_label_kind_inventory_frozen():
+    # the 12 kinds the reference supports (main.go:115-379)
+    assert LABEL_KINDS == sorted([
+        "firmware",
+        "family",
+        "driver-version",
+        "driver-src-version",
+        "device-id",
+        "product-name",
+        "vram",
+        "simd-count",
+        "cu-count",
+        "compute-memory-partition",
+        "compute-partitioning-supported",
+        "memory-partitioning-supported",
+    ])
+
+
+def test_label_prefixes():
+    assert create_label_prefix("vram") == "amd.com/gpu.vram"
+    assert create_label_prefix("vram", True) == "beta.amd.com/gpu.vram"
+
+
+def test_generate_labels_mi355x(fake_mi355x_8):
+    enabled = {k: True for k in LABEL_KINDS}
+    labels = generate_labels(enabled, fake_mi355x_8.paths)
+    # gfx950 values: 288G HBM3E, 256 CUs, 1024 SIMDs
+    assert labels["amd.com/gpu.vram"] == "288G"
+    assert labels["beta.amd.com/gpu.vram"] == "288G"
+    assert labels["beta.amd.com/gpu.vram.288G"] == "8"
+    assert labels["amd.com/gpu.cu-count"] == "256"
+    assert labels["amd.com/gpu.simd-count"] == "1024"
+    assert labels["amd.com/gpu.device-id"] == "75a3"
+    assert labels["amd.com/gpu.product-name"] == "AMD_Instinct_MI355_OAM"
+    assert labels["amd.com/gpu.driver-version"] == "6.14.14"
+    assert labels["amd.com/gpu.driver-src-version"] == "FAKE123456789"
+    assert labels["amd.com/gpu.compute-memory-partition"] == "spx_nps1"
+    assert labels["amd.com/gpu.compute-partitioning-supported"] == "true"
+    assert labels["amd.com/gpu.memory-partitioning-supported"] == "true"
+    # family via sysfs fallback (no ioctl on a fake tree): gfx950 -> AI
+    assert labels["amd.com/gpu.family"] == "AI"
+    assert labels["beta.amd.com/gpu.family.AI"] == "8"
+
+
+def test_generate_labels_subset(fake_mi355x_8):
+    labels = generate_labels({"vram": True}, fake_mi355x_8.paths)
+    assert set(labels) == {
+        "amd.com/gpu.vram",
+        "beta.amd.com/gpu.vram",
+        "beta.amd.com/gpu.vram.288G",
+    }
+
+
+def test_remove_old_node_labels():
+    labels = {
+        "amd.com/gpu.vram": "288G",
+        "beta.amd.com/gpu.family": "AI",
+        "beta.amd.com/gpu.family.AI": "8",
+        "beta.amd.com/gpu.firmware.MEC.fw.177": "8",
+        "amd.com/gpu.device-id.75a3": "8",
+        "kubernetes.io/hostname": "n0",       # unrelated: preserved
+        "amd.com/other": "x",                 # not ours: preserved
+    }
+    remove_old_node_labels(labels)
+    assert labels == {"kubernetes.io/hostname": "n0", "amd.com/other": "x"}
+
+
+def test_product_name_ids_lookup():
+    assert product_name_from_ids(0x75A3) == "AMD Instinct MI355 OAM"
+    assert product_name_from_ids(0x75A3, 0x00) == "AMD Instinct MI355 OAM"
+    assert product_name_from_ids(0x740F, 0xC1) == "AMD Instinct MI210"
+    assert product_name_from_ids(0xDEAD) is None
+
+
+def test_reconcile_against_fake_api(fake_mi355x_8):
+    fake = FakeK8s(
+        node_name="mi355x-node-0",
+        initial_labels={
+            "kubernetes.io/hostname": "mi355x-node-0",
+            "amd.com/gpu.vram": "192G",            # stale: must be replaced
+            "beta.amd.com/gpu.family": "OLD",      # stale + counter
+            "beta.amd.com/gpu.family.OLD": "4",
+        },
+    ).start()
+    try:
+        labels = generate_labels({k: True for k in LABEL_KINDS},
+                                 fake_mi355x_8.paths)
+        client = K8sClient(base_url=fake.base_url)
+        ctl = NodeLabelController(client, "mi355x-node-0", labels)
+        patch = ctl.reconcile()
+
+        assert patch["beta.amd.com/gpu.family.OLD"] is None
+        assert fake.labels["amd.com/gpu.vram"] == "288G"
+        assert fake.labels["beta.amd.com/gpu.family"] == "AI"
+        assert "beta.amd.com/gpu.family.OLD" not in fake.labels
+        assert fake.labels["kubernetes.io/hostname"] == "mi355x-node-0"
+
+        # second reconcile is a no-op
+        assert ctl.reconcile() == {}
+    finally:
+        fake.stop()
+
+
+def test_watch_triggers_reconcile(fake_mi355x_8):
+    fake = FakeK8s(node_name="n1").start()
+    try:
+        labels = generate_labels({"vram": True}, fake_mi355x_8.paths)
+        client = K8sClient(base_url=fake.base_url)
+        ctl = NodeLabelController(client, "n1", labels)
+        ctl.reconcile()
+        # wipe labels behind the controller's back, then fire an ADDED event
+        fake.labels.clear()
+        ctl.run(block=False)
+        fake.push_event("ADDED")
+        import time
+
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline and "amd.com/gpu.vram" not in fake.labels:
+            time.sleep(0.05)
+        assert fake.labels.get("amd.com/gpu.vram") == "288G"
+        ctl.stop()
+    finally:
+        fake.stop()
+
+
+def test_cli_labeller_oneshot(fake_mi355x_8, monkeypatch):
+    fake = FakeK8s(node_name="cli-node").start()
+    try:
+        monkeypatch.setenv("DS_NODE_NAME", "cli-node")
+        from k8s_device_plugin_amd.cli import labeller_main
+
+        rc = labeller_main([
+            "--vram", "--cu_count", "--family",
+            "--sysroot", fake_mi355x_8.paths.root,
+            "--api-server", fake.base_url,
+            "--oneshot",
+        ])
+        assert rc == 0
+        assert fake.labels["amd.com/gpu.vram"] == "288G"
+        assert fake.labels["amd.com/gpu.cu-count"] == "256"
+    finally:
+        fake.stop()
