@@ -174,9 +174,17 @@ class BestEffortPolicy:
                 queue.append(s)
 
         # breadth-first extension across GPUs for requests no single GPU
-        # can satisfy (reference: device.go:406-441)
-        while queue:
-            cur = queue.pop(0)
+        # can satisfy (reference: device.go:406-441).  An incomplete queue
+        # state always holds ALL free devices of each parent group, so its
+        # id set — and total weight — is fully determined by the parent
+        # SET; the reference re-expands every parent-order permutation
+        # (O(G!) states), we dedupe on the set (O(2^G)), which makes
+        # 64-partition CPX requests tractable without changing any result.
+        seen_parent_sets = {s.parents for s in queue}
+        qi = 0
+        while qi < len(queue):
+            cur = queue[qi]
+            qi += 1
             if len(cur.parents) == len(groups):
                 continue
             for idx, g in enumerate(groups):
@@ -190,6 +198,7 @@ class BestEffortPolicy:
                         final.append(finish(s))
                         done = True
                         break
-                if not done:
+                if not done and s.parents not in seen_parent_sets:
+                    seen_parent_sets.add(s.parents)
                     queue.append(s)
         return final

@@ -128,21 +128,30 @@ def _sanitize_name(name: str) -> str:
 # ---- generators: each (devices, paths, topo) -> {label: value} ----
 
 def _gen_firmware(devices, paths, topo) -> Dict[str, str]:
+    from ..topology.firmware import debugfs_firmware_path, parse_debugfs_firmware
+
     drm = load_drmctl()
-    if drm is None:
-        return {}
     counts: Dict[str, int] = {}
     for d in _physical_gpus(devices):
         if not d.kfd_backed:
             continue
-        try:
-            fw = drm.query_firmware(f"/dev/dri/renderD{d.render_d}")
-        except RuntimeError as e:
-            log.error("firmware query failed for %s: %s", d.id, e)
+        feat = fwv = None
+        if drm is not None:
+            try:
+                fw = drm.query_firmware(f"/dev/dri/renderD{d.render_d}")
+                feat, fwv = fw["feature"], fw["firmware"]
+            except RuntimeError as e:
+                log.error("firmware ioctl failed for %s: %s", d.id, e)
+        if feat is None:
+            # debugfs fallback (root-only; present in privileged DaemonSets)
+            feat, fwv = parse_debugfs_firmware(
+                debugfs_firmware_path(d.card, os.path.join(paths.root, "sys/kernel/debug"))
+            )
+        if not feat:
             continue
-        for blk, ver in fw["feature"].items():
+        for blk, ver in feat.items():
             counts[f"{blk}.feat.{ver}"] = counts.get(f"{blk}.feat.{ver}", 0) + 1
-        for blk, ver in fw["firmware"].items():
+        for blk, ver in fwv.items():
             counts[f"{blk}.fw.{ver}"] = counts.get(f"{blk}.fw.{ver}", 0) + 1
     # firmware labels exist only in the experimental namespace
     # (reference: main.go:137-142)

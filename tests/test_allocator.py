@@ -173,3 +173,27 @@ def test_init_requires_links(tmp_path):
         policy.init(devices.values(), topology=topo)
     with pytest.raises(AllocationError):
         policy.init([], topology=topo)
+
+
+def test_cpx_large_requests_fast(fake_mi355x_cpx):
+    """30/56/60-device requests on a 64-partition node must stay
+    interactive (the BFS dedupes parent sets; the reference's
+    permutation expansion would be O(G!))."""
+    import time
+
+    policy, devices = make_policy(fake_mi355x_cpx)
+    ids = sorted(devices)
+    for size, expect_gpus in ((30, 4), (56, 7), (60, 8)):
+        t0 = time.perf_counter()
+        out = policy.allocate(ids, [], size)
+        dt = time.perf_counter() - t0
+        assert len(out) == size
+        assert len({devices[i].dev_id for i in out}) == expect_gpus
+        assert dt < 2.0, f"size={size} took {dt:.2f}s"
+
+
+def test_cpx_allocate_all(fake_mi355x_cpx):
+    policy, devices = make_policy(fake_mi355x_cpx)
+    ids = sorted(devices)
+    out = policy.allocate(ids, [], 64)  # fast path: available == size
+    assert len(out) == 64
