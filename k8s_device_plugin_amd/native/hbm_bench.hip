@@ -117,7 +117,7 @@ int main(int argc, char **argv) {
 
     const int iters = 10;
     struct { const char *name; double gbps; } best{"", 0};
-    for (int blocks : {2048, 4096, 8192, 16384}) {
+    for (int blocks : {8192, 16384, 32768, 65536, 131072}) {
         double p = bench(copy_plain, src, dst, n, blocks, 256, iters);
         double u = bench(copy_unroll4, src, dst, n, blocks, 256, iters);
         double t = bench(copy_nt, src, dst, n, blocks, 256, iters);

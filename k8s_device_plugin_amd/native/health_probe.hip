@@ -103,11 +103,29 @@ __global__ void lds_probe_kernel(uint32_t *err_count, int lds_words) {
 
 // ---------------- HBM streaming copy ----------------
 
-__global__ void hbm_copy_kernel(const float4 *__restrict__ src,
-                                float4 *__restrict__ dst, size_t n) {
-    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+// Nontemporal unrolled float4 stream: measured fastest variant on MI355X
+// (profiles/r01_bench_mi355x.md; nt hints keep the one-shot stream out of
+// L1/L2, 4 independent loads per thread hide HBM latency).
+using f32x4_nt = __attribute__((ext_vector_type(4))) float;
+
+__global__ void hbm_copy_kernel(const float4 *__restrict__ src4,
+                                float4 *__restrict__ dst4, size_t n) {
+    const f32x4_nt *__restrict__ src = reinterpret_cast<const f32x4_nt *>(src4);
+    f32x4_nt *__restrict__ dst = reinterpret_cast<f32x4_nt *>(dst4);
     size_t stride = (size_t)gridDim.x * blockDim.x;
-    for (; i < n; i += stride) dst[i] = src[i];
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + 3 * stride < n; i += 4 * stride) {
+        f32x4_nt a = __builtin_nontemporal_load(&src[i]);
+        f32x4_nt b = __builtin_nontemporal_load(&src[i + stride]);
+        f32x4_nt c = __builtin_nontemporal_load(&src[i + 2 * stride]);
+        f32x4_nt d = __builtin_nontemporal_load(&src[i + 3 * stride]);
+        __builtin_nontemporal_store(a, &dst[i]);
+        __builtin_nontemporal_store(b, &dst[i + stride]);
+        __builtin_nontemporal_store(c, &dst[i + 2 * stride]);
+        __builtin_nontemporal_store(d, &dst[i + 3 * stride]);
+    }
+    for (; i < n; i += stride)
+        __builtin_nontemporal_store(__builtin_nontemporal_load(&src[i]), &dst[i]);
 }
 
 __global__ void fill_pattern_kernel(float4 *buf, size_t n) {
@@ -194,8 +212,9 @@ py::dict run_probe(int device, size_t hbm_bytes) {
         float4 *src, *dst;
         HIP_CHECK(hipMalloc(&src, n * sizeof(float4)));
         HIP_CHECK(hipMalloc(&dst, n * sizeof(float4)));
-        // >>256 workgroups to fill all 8 XCDs (256 CUs)
-        const int blocks = 8192, threads = 256, iters = 5;
+        // >>256 workgroups to fill all 8 XCDs (256 CUs); 16k blocks was the
+        // measured sweet spot for the nt stream (gpurun_out/hbm_variants.txt)
+        const int blocks = 16384, threads = 256, iters = 5;
         hipLaunchKernelGGL(fill_pattern_kernel, dim3(blocks), dim3(threads), 0, 0,
                            src, n);
         hipLaunchKernelGGL(hbm_copy_kernel, dim3(blocks), dim3(threads), 0, 0,
