@@ -128,3 +128,25 @@ def test_bench_single_gpu():
     assert rec["config"]["present_gpus"] >= 1
     assert rec["config"]["advertised_gpus"] >= 1
     assert rec["value"] > 0
+
+
+def test_firmware_ioctl_vs_debugfs(live_devices):
+    """Cross-check ioctl firmware versions against debugfs when readable
+    (reference: amdgpu_test.go:45-75)."""
+    from k8s_device_plugin_amd.native import load_drmctl
+    from k8s_device_plugin_amd.topology.firmware import (
+        debugfs_firmware_path,
+        parse_debugfs_firmware,
+    )
+
+    _, devs = live_devices
+    drm = load_drmctl(required=True)
+    d = sorted((x for x in devs.values() if x.kfd_backed),
+               key=lambda x: x.render_d)[0]
+    fw = drm.query_firmware(f"/dev/dri/renderD{d.render_d}")
+    feat_dbg, fw_dbg = parse_debugfs_firmware(debugfs_firmware_path(d.card))
+    if not fw_dbg:
+        pytest.skip("debugfs not readable in this container")
+    for blk in ("MEC", "RLC", "SDMA0"):
+        if blk in fw_dbg and fw["firmware"][blk]:
+            assert fw["firmware"][blk] == fw_dbg[blk], blk

@@ -219,3 +219,22 @@ def test_heterogeneous_bucketing(tmp_path):
     finally:
         mgr.stop()
         kubelet.stop()
+
+
+def test_two_streams_both_get_heartbeats(harness):
+    """kubelet reconnects can leave two ListAndWatch streams briefly open;
+    every stream must receive health refreshes."""
+    kubelet, mgr, _ = harness
+    mgr.run(["gpu"])
+    reg = kubelet.wait_for_registration()
+    stub = kubelet.connect(reg.endpoint)
+    call1, q1 = open_stream(stub)
+    call2, q2 = open_stream(stub)
+    q1.get(timeout=5)
+    q2.get(timeout=5)
+    mgr.heartbeat_all()
+    r1 = q1.get(timeout=10)
+    r2 = q2.get(timeout=10)
+    assert len(r1.devices) == len(r2.devices) == 8
+    call1.cancel()
+    call2.cancel()
