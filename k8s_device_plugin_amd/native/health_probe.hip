@@ -212,9 +212,12 @@ py::dict run_probe(int device, size_t hbm_bytes) {
         float4 *src, *dst;
         HIP_CHECK(hipMalloc(&src, n * sizeof(float4)));
         HIP_CHECK(hipMalloc(&dst, n * sizeof(float4)));
-        // >>256 workgroups to fill all 8 XCDs (256 CUs); 16k blocks was the
-        // measured sweet spot for the nt stream (gpurun_out/hbm_variants.txt)
-        const int blocks = 16384, threads = 256, iters = 5;
+        // >>256 workgroups to fill all 8 XCDs (256 CUs); bandwidth keeps
+        // climbing with grid size for the nt stream (measured sweep in
+        // profiles/), so give each thread ~4 unrolled iterations
+        const int threads = 256, iters = 5;
+        long want = (long)(n / (threads * 4));
+        const int blocks = (int)(want < 1024 ? 1024 : (want > 131072 ? 131072 : want));
         hipLaunchKernelGGL(fill_pattern_kernel, dim3(blocks), dim3(threads), 0, 0,
                            src, n);
         hipLaunchKernelGGL(hbm_copy_kernel, dim3(blocks), dim3(threads), 0, 0,
