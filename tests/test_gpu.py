@@ -150,3 +150,26 @@ def test_firmware_ioctl_vs_debugfs(live_devices):
     for blk in ("MEC", "RLC", "SDMA0"):
         if blk in fw_dbg and fw["firmware"][blk]:
             assert fw["firmware"][blk] == fw_dbg[blk], blk
+
+
+def test_allocate_paths_are_real_devices(live_devices):
+    """The device nodes Allocate returns must exist and answer ioctls —
+    validates the full path kubelet would inject into a container."""
+    from k8s_device_plugin_amd.native import load_drmctl
+    from k8s_device_plugin_amd.plugin import AMDGPUPlugin
+    from k8s_device_plugin_amd.protos import deviceplugin as dp
+
+    paths, devs = live_devices
+    drm = load_drmctl(required=True)
+    plugin = AMDGPUPlugin(resource="gpu", paths=paths)
+    plugin.start()
+    backed = sorted((d for d in devs.values() if d.kfd_backed),
+                    key=lambda d: d.id)
+    req = dp.AllocateRequest()
+    req.container_requests.add().devices_ids.append(backed[0].id)
+    resp = plugin.Allocate(req, None)
+    specs = [s.host_path for s in resp.container_responses[0].devices]
+    assert "/dev/kfd" in specs
+    render = [p for p in specs if "renderD" in p]
+    assert render and os.path.exists(render[0])
+    assert drm.dev_functional(render[0])
