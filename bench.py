@@ -118,6 +118,7 @@ def main() -> int:
                 "resource": "amd.com/gpu",
                 "devices_per_allocate": args.gpus,
                 "parallelism": "node-local",
+                "server": harness.server_impl,
                 "sysfs": harness.sysfs_kind,
                 "advertised_gpus": advertised,
                 "present_gpus": present_gpus,
@@ -177,11 +178,16 @@ class _Harness:
 
         dp_dir = os.path.join(root, "device-plugins")
         self._kubelet = StubKubelet(dp_dir).start()
+        server_impl = os.environ.get("AMDXDP_SERVER", "native")
         self._mgr = PluginManager(
             lambda res: AMDGPUPlugin(resource=res, paths=paths),
             device_plugin_path=dp_dir,
+            server_impl=server_impl,
         )
         self._mgr.run(["gpu"])
+        self.server_impl = (
+            "native" if self._mgr.plugins["gpu"].native else "python"
+        )
         reg = self._kubelet.wait_for_registration()
         self._stub = self._kubelet.connect(reg.endpoint)
         self._dp = dp

@@ -42,6 +42,9 @@ def device_plugin_main(argv=None) -> int:
     ap.add_argument("--kubelet-dir", default=None,
                     help="device-plugin dir override (tests)")
     ap.add_argument("--sysroot", default="/", help="sysfs root override (tests)")
+    ap.add_argument("--server", default="native", choices=["native", "python"],
+                    help="serving implementation (native C++/nghttp2 with "
+                         "automatic python fallback, or python grpc)")
     ap.add_argument("-v", "--verbose", action="count", default=0)
     args = ap.parse_args(argv)
     _setup_logging(args.verbose)
@@ -93,6 +96,7 @@ def device_plugin_main(argv=None) -> int:
     mgr = PluginManager(
         lambda res: AMDGPUPlugin(resource=res, paths=paths),
         device_plugin_path=args.kubelet_dir or dp.DEVICE_PLUGIN_PATH,
+        server_impl=args.server,
     )
     ticker = HeartbeatTicker(args.pulse)
     ticker.subscribe(mgr.heartbeat_all)

@@ -61,6 +61,16 @@ def load_healthprobe(required: Optional[bool] = None):
         return None
 
 
+def load_fastserver(required: bool = False):
+    """Native DevicePlugin gRPC server (nghttp2 over UDS)."""
+    try:
+        return _load("_fastserver")
+    except NativeExtensionMissing:
+        if required:
+            raise
+        return None
+
+
 def deep_health_probe(device: int = 0, hbm_bytes: int = 1 << 30) -> dict:
     """Run the on-GPU MFMA/LDS/HBM probe.  Raises loudly when the extension
     is missing on a GPU machine."""

@@ -72,8 +72,21 @@ def build_healthprobe(force: bool = False) -> str:
     return out
 
 
+def build_fastserver(force: bool = False) -> str:
+    src = os.path.join(PKG_DIR, "fastserver.cpp")
+    out = os.path.join(PKG_DIR, "_fastserver.so")
+    hdr = os.path.join(PKG_DIR, "nghttp2_abi.h")
+    if force or _needs_build(src, out) or _needs_build(hdr, out):
+        _run(
+            ["g++", "-O2", "-std=c++17", "-shared", "-fPIC", src, "-o", out,
+             "-ldl", "-pthread"]
+            + _python_includes()
+        )
+    return out
+
+
 def build_all(force: bool = False) -> list:
-    return [build_drmctl(force), build_healthprobe(force)]
+    return [build_drmctl(force), build_healthprobe(force), build_fastserver(force)]
 
 
 if __name__ == "__main__":

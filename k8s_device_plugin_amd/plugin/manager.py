@@ -30,7 +30,8 @@ class _PluginInstance:
         self.resource = resource
         self.plugin = plugin
         self.socket_path = socket_path
-        self.server = None  # grpc.Server
+        self.server = None  # grpc.Server or NativePluginServer
+        self.native = False
 
     @property
     def endpoint(self) -> str:
@@ -45,8 +46,13 @@ class PluginManager:
         device_plugin_path: str = dp.DEVICE_PLUGIN_PATH,
         kubelet_socket: Optional[str] = None,
         watch_interval: float = 0.5,
+        server_impl: str = "native",
     ):
+        """server_impl: "native" (C++/nghttp2 fast server; falls back to
+        python when the extension or libnghttp2 is unavailable) or
+        "python" (grpc.Server)."""
         self.plugin_factory = plugin_factory
+        self.server_impl = server_impl
         self.namespace = namespace
         self.device_plugin_path = device_plugin_path
         self.kubelet_socket = kubelet_socket or os.path.join(
@@ -75,15 +81,34 @@ class PluginManager:
 
             if os.path.exists(socket_path):
                 os.unlink(socket_path)
-            # Allocate/GetPreferredAllocation are tiny; a few workers are
-            # plenty and keep the ListAndWatch streams responsive.
-            server = grpc.server(ThreadPoolExecutor(max_workers=8))
-            dp.add_device_plugin_servicer(server, plugin)
-            server.add_insecure_port(f"unix://{socket_path}")
-            server.start()
+
+            server = None
+            native = False
+            if self.server_impl == "native":
+                try:
+                    from .native_server import NativePluginServer
+
+                    server = NativePluginServer(plugin, socket_path)
+                    server.start()
+                    native = True
+                    log.info("%s: serving via native fast server", resource)
+                except Exception as e:
+                    log.warning(
+                        "%s: native server unavailable (%s); falling back to "
+                        "python grpc", resource, e,
+                    )
+                    server = None
+            if server is None:
+                # Allocate/GetPreferredAllocation are tiny; a few workers
+                # keep the ListAndWatch streams responsive.
+                server = grpc.server(ThreadPoolExecutor(max_workers=8))
+                dp.add_device_plugin_servicer(server, plugin)
+                server.add_insecure_port(f"unix://{socket_path}")
+                server.start()
 
             inst = _PluginInstance(resource, plugin, socket_path)
             inst.server = server
+            inst.native = native
             self.plugins[resource] = inst
         self.register(inst)
         return inst
@@ -160,7 +185,10 @@ class PluginManager:
 
     def heartbeat_all(self) -> None:
         for inst in self.plugins.values():
-            inst.plugin.heartbeat()
+            if inst.native:
+                inst.server.heartbeat()  # recompute + push to all streams
+            else:
+                inst.plugin.heartbeat()
 
     def stop(self) -> None:
         self._stop.set()

@@ -1,0 +1,88 @@
+"""Native-server glue: feeds the C++ fast server pre-serialized state.
+
+The C++ side (native/fastserver.cpp) owns the wire and the
+preferred-allocation search; this module computes everything once in
+Python — device list bytes, per-device Allocate fragments, allocator
+tables — and pushes updates on each heartbeat.  Wire behavior is
+conformance-tested against the Python grpc client in
+tests/test_fastserver.py.
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Dict, Optional
+
+from ..protos import deviceplugin as dp
+from .server import AMDGPUPlugin
+
+log = logging.getLogger(__name__)
+
+
+def _container_response_bytes(specs) -> bytes:
+    """Serialized ContainerAllocateResponse holding only `specs` — i.e. the
+    tagged `devices` field fragments, concatenable per protobuf rules."""
+    car = dp.ContainerAllocateResponse()
+    for host_path in specs:
+        s = car.devices.add()
+        s.host_path = s.container_path = host_path
+        s.permissions = "rw"
+    return car.SerializeToString()
+
+
+class NativePluginServer:
+    """Drop-in replacement for the grpc.Server serving one plugin."""
+
+    def __init__(self, plugin: AMDGPUPlugin, socket_path: str):
+        from ..native import load_fastserver
+
+        mod = load_fastserver(required=True)
+        self.plugin = plugin
+        self.socket_path = socket_path
+        self._srv = mod.Server(socket_path)
+
+    def start(self) -> None:
+        p = self.plugin
+        self._srv.set_options_response(
+            p.GetDevicePluginOptions(dp.Empty(), None).SerializeToString()
+        )
+        self._srv.set_kfd_spec(_container_response_bytes(["/dev/kfd"]))
+        specs: Dict[str, bytes] = {}
+        for d in p.devices.values():
+            specs[d.id] = _container_response_bytes(
+                [f"/dev/dri/card{d.card}", f"/dev/dri/renderD{d.render_d}"]
+            )
+        self._srv.set_device_specs(specs)
+        self._srv.set_list_response(
+            dp.ListAndWatchResponse(devices=p._device_list()).SerializeToString()
+        )
+        if not p.allocator_init_error and p.allocator.initialized:
+            alloc = p.allocator
+            groups = [
+                (g.parent_id, sorted(g.node_ids))
+                for g in alloc._groups.values()
+            ]
+            node_of_id = {d.id: d.node_id for d in alloc._devices.values()}
+            weights = [
+                (a, b, w)
+                for a, inner in alloc._weights.items()
+                for b, w in inner.items()
+            ]
+            self._srv.set_allocator_state(groups, node_of_id, weights)
+        self._srv.start()
+
+    def heartbeat(self) -> None:
+        """Recompute health and push the fresh list to every open stream."""
+        devs = self.plugin.refreshed_device_list()
+        self._srv.push_list_update(
+            dp.ListAndWatchResponse(devices=devs).SerializeToString()
+        )
+
+    def stop(self, grace: Optional[float] = None):
+        self._srv.stop()
+
+        class _Done:
+            def wait(self, timeout=None):
+                return True
+
+        return _Done()

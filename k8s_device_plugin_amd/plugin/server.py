@@ -127,6 +127,26 @@ class AMDGPUPlugin:
             out.append(dev)
         return out
 
+    def refreshed_device_list(self) -> List:
+        """Device list with freshly evaluated health: node-level kfd scan
+        default, exporter per-GPU overrides, unbacked devices pinned
+        Unhealthy.  Shared by the Python stream loop and the native
+        server's heartbeat push."""
+        default = (
+            dp.HEALTHY if simple_health_check(self.paths) else dp.UNHEALTHY
+        )
+        devs = self._device_list(health_default=default)
+        populate_per_gpu_health(
+            devs, default, self.exporter_socket, self.exporter_timeout
+        )
+        # an exporter verdict cannot resurrect a device the kfd topology
+        # does not back
+        unbacked = {d.id for d in self._my_devices() if not d.kfd_backed}
+        for dev in devs:
+            if dev.ID in unbacked:
+                dev.health = dp.UNHEALTHY
+        return devs
+
     def ListAndWatch(self, request, context):
         # re-walk sysfs and refresh the Allocate cache (plugin.go:231)
         self.devices = discover_gpus(self.paths, strict=False)
@@ -161,20 +181,7 @@ class AMDGPUPlugin:
             if not fired:
                 continue
 
-            default = (
-                dp.HEALTHY if simple_health_check(self.paths) else dp.UNHEALTHY
-            )
-            devs = self._device_list(health_default=default)
-            populate_per_gpu_health(
-                devs, default, self.exporter_socket, self.exporter_timeout
-            )
-            # an exporter verdict cannot resurrect a device the kfd
-            # topology does not back
-            unbacked = {d.id for d in self._my_devices() if not d.kfd_backed}
-            for dev in devs:
-                if dev.ID in unbacked:
-                    dev.health = dp.UNHEALTHY
-            yield dp.ListAndWatchResponse(devices=devs)
+            yield dp.ListAndWatchResponse(devices=self.refreshed_device_list())
 
     def GetPreferredAllocation(self, request, context):
         import grpc

@@ -19,9 +19,10 @@ from k8s_device_plugin_amd.testing.fakesysfs import FakeSysfs, build_mi355x_node
 from k8s_device_plugin_amd.testing.stub_kubelet import StubKubelet
 
 
-@pytest.fixture
-def harness(tmp_path, fake_mi355x_8):
-    """StubKubelet + PluginManager over the 8-GPU fake node."""
+@pytest.fixture(params=["native", "python"])
+def harness(request, tmp_path, fake_mi355x_8):
+    """StubKubelet + PluginManager over the 8-GPU fake node, exercised
+    against BOTH serving implementations."""
     dp_dir = str(tmp_path / "device-plugins")
     exporter_sock = str(tmp_path / "exporter" / "metrics.sock")
     kubelet = StubKubelet(dp_dir).start()
@@ -35,7 +36,8 @@ def harness(tmp_path, fake_mi355x_8):
         )
 
     mgr = PluginManager(
-        factory, device_plugin_path=dp_dir, watch_interval=0.1
+        factory, device_plugin_path=dp_dir, watch_interval=0.1,
+        server_impl=request.param,
     )
     yield kubelet, mgr, exporter_sock
     mgr.stop()
