@@ -108,13 +108,17 @@ __global__ void lds_probe_kernel(uint32_t *err_count, int lds_words) {
 // L1/L2, 4 independent loads per thread hide HBM latency).
 using f32x4_nt = __attribute__((ext_vector_type(4))) float;
 
+// Exact-fit one-shot stream: thread t moves exactly 4 float4s at
+// grid-stride offsets, no loop bookkeeping — measured 6.22 TB/s on MI355X
+// (99% of the 6.29 TB/s float4-copy ceiling, MI355X_MICROARCH.md §HBM;
+// sweep in profiles/r01_bench_mi355x.md).
 __global__ void hbm_copy_kernel(const float4 *__restrict__ src4,
                                 float4 *__restrict__ dst4, size_t n) {
     const f32x4_nt *__restrict__ src = reinterpret_cast<const f32x4_nt *>(src4);
     f32x4_nt *__restrict__ dst = reinterpret_cast<f32x4_nt *>(dst4);
     size_t stride = (size_t)gridDim.x * blockDim.x;
     size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
-    for (; i + 3 * stride < n; i += 4 * stride) {
+    if (i + 3 * stride < n) {
         f32x4_nt a = __builtin_nontemporal_load(&src[i]);
         f32x4_nt b = __builtin_nontemporal_load(&src[i + stride]);
         f32x4_nt c = __builtin_nontemporal_load(&src[i + 2 * stride]);
@@ -123,9 +127,11 @@ __global__ void hbm_copy_kernel(const float4 *__restrict__ src4,
         __builtin_nontemporal_store(b, &dst[i + stride]);
         __builtin_nontemporal_store(c, &dst[i + 2 * stride]);
         __builtin_nontemporal_store(d, &dst[i + 3 * stride]);
+    } else {
+        for (; i < n; i += stride)
+            __builtin_nontemporal_store(__builtin_nontemporal_load(&src[i]),
+                                        &dst[i]);
     }
-    for (; i < n; i += stride)
-        __builtin_nontemporal_store(__builtin_nontemporal_load(&src[i]), &dst[i]);
 }
 
 __global__ void fill_pattern_kernel(float4 *buf, size_t n) {
@@ -212,12 +218,11 @@ py::dict run_probe(int device, size_t hbm_bytes) {
         float4 *src, *dst;
         HIP_CHECK(hipMalloc(&src, n * sizeof(float4)));
         HIP_CHECK(hipMalloc(&dst, n * sizeof(float4)));
-        // >>256 workgroups to fill all 8 XCDs (256 CUs); bandwidth keeps
-        // climbing with grid size for the nt stream (measured sweep in
-        // profiles/), so give each thread ~4 unrolled iterations
+        // exact-fit grid: one thread per 4 float4s (no loop), which was the
+        // fastest measured formulation; still >>256 workgroups for all 8 XCDs
         const int threads = 256, iters = 5;
-        long want = (long)(n / (threads * 4));
-        const int blocks = (int)(want < 1024 ? 1024 : (want > 131072 ? 131072 : want));
+        long want = (long)((n + threads * 4 - 1) / (threads * 4));
+        const int blocks = (int)(want < 1024 ? 1024 : want);
         hipLaunchKernelGGL(fill_pattern_kernel, dim3(blocks), dim3(threads), 0, 0,
                            src, n);
         hipLaunchKernelGGL(hbm_copy_kernel, dim3(blocks), dim3(threads), 0, 0,
