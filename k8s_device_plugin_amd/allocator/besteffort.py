@@ -91,6 +91,17 @@ class BestEffortPolicy:
     def initialized(self) -> bool:
         return bool(self._weights)
 
+    def export_state(self):
+        """State for the native fast server's in-C++ search: (groups as
+        [(parent_id, sorted node_ids)], {device_id: node_id},
+        [(node_a, node_b, weight)])."""
+        groups = [(g.parent_id, sorted(g.node_ids)) for g in self._groups.values()]
+        node_of_id = {d.id: d.node_id for d in self._devices.values()}
+        weights = [
+            (a, b, w) for a, inner in self._weights.items() for b, w in inner.items()
+        ]
+        return groups, node_of_id, weights
+
     def allocate(
         self,
         available_ids: Sequence[str],

@@ -57,17 +57,7 @@ class NativePluginServer:
             dp.ListAndWatchResponse(devices=p._device_list()).SerializeToString()
         )
         if not p.allocator_init_error and p.allocator.initialized:
-            alloc = p.allocator
-            groups = [
-                (g.parent_id, sorted(g.node_ids))
-                for g in alloc._groups.values()
-            ]
-            node_of_id = {d.id: d.node_id for d in alloc._devices.values()}
-            weights = [
-                (a, b, w)
-                for a, inner in alloc._weights.items()
-                for b, w in inner.items()
-            ]
+            groups, node_of_id, weights = p.allocator.export_state()
             self._srv.set_allocator_state(groups, node_of_id, weights)
         self._srv.start()
 

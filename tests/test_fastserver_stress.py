@@ -1,0 +1,138 @@
+"""Native server stress/robustness: concurrent clients, stream churn,
+mid-stream disconnects, malformed frames."""
+
+import queue
+import socket
+import threading
+import time
+
+import grpc
+import pytest
+
+from k8s_device_plugin_amd.plugin import AMDGPUPlugin
+from k8s_device_plugin_amd.plugin.native_server import NativePluginServer
+from k8s_device_plugin_amd.protos import deviceplugin as dp
+
+
+@pytest.fixture
+def native(tmp_path, fake_mi355x_8):
+    sock = str(tmp_path / "n.sock")
+    plugin = AMDGPUPlugin(resource="gpu", paths=fake_mi355x_8.paths)
+    plugin.start()
+    srv = NativePluginServer(plugin, sock)
+    srv.start()
+    yield plugin, srv, sock
+    srv.stop()
+
+
+def test_concurrent_allocate_many_channels(native):
+    plugin, _, sock = native
+    ids = sorted(plugin.devices)
+    errors = []
+
+    def worker(n):
+        try:
+            ch = grpc.insecure_channel(f"unix://{sock}")
+            stub = dp.DevicePluginStub(ch)
+            req = dp.AllocateRequest()
+            req.container_requests.add().devices_ids.extend(ids[: 1 + n % 4])
+            for _ in range(100):
+                resp = stub.Allocate(req, timeout=10)
+                assert len(resp.container_responses[0].devices) == 1 + 2 * (1 + n % 4)
+            ch.close()
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    threads = [threading.Thread(target=worker, args=(i,)) for i in range(8)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+    assert not errors, errors
+
+
+def test_stream_churn(native):
+    """Open/cancel ListAndWatch repeatedly while heartbeats push."""
+    plugin, srv, sock = native
+    ch = grpc.insecure_channel(f"unix://{sock}")
+    stub = dp.DevicePluginStub(ch)
+    stop = threading.Event()
+
+    def beater():
+        while not stop.is_set():
+            srv.heartbeat()
+            time.sleep(0.005)
+
+    t = threading.Thread(target=beater, daemon=True)
+    t.start()
+    try:
+        for _ in range(30):
+            call = stub.ListAndWatch(dp.Empty())
+            it = iter(call)
+            first = next(it)
+            assert len(first.devices) == 8
+            call.cancel()
+        # a long-lived stream still works after the churn
+        call = stub.ListAndWatch(dp.Empty())
+        it = iter(call)
+        next(it)
+        srv.heartbeat()
+        got = next(it)
+        assert len(got.devices) == 8
+        call.cancel()
+    finally:
+        stop.set()
+        t.join(timeout=5)
+    ch.close()
+
+
+def test_abrupt_disconnect_mid_stream(native):
+    """Kill the TCP-level connection while a stream is open; the server
+    must keep serving other clients."""
+    plugin, srv, sock = native
+    ch1 = grpc.insecure_channel(f"unix://{sock}")
+    stub1 = dp.DevicePluginStub(ch1)
+    call = stub1.ListAndWatch(dp.Empty())
+    next(iter(call))
+    # abrupt close without goaway
+    ch1.close()
+    time.sleep(0.1)
+
+    ch2 = grpc.insecure_channel(f"unix://{sock}")
+    stub2 = dp.DevicePluginStub(ch2)
+    opts = stub2.GetDevicePluginOptions(dp.Empty(), timeout=5)
+    assert opts.get_preferred_allocation_available
+    ch2.close()
+
+
+def test_garbage_bytes_do_not_kill_server(native):
+    plugin, srv, sock = native
+    s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    s.connect(sock)
+    s.sendall(b"GET / HTTP/1.1\r\nHost: x\r\n\r\n")  # not http/2
+    time.sleep(0.1)
+    s.close()
+
+    s2 = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    s2.connect(sock)
+    s2.sendall(b"\x00" * 64)
+    s2.close()
+
+    ch = grpc.insecure_channel(f"unix://{sock}")
+    stub = dp.DevicePluginStub(ch)
+    assert stub.GetDevicePluginOptions(dp.Empty(), timeout=5) is not None
+    ch.close()
+
+
+def test_restart_server_same_socket(tmp_path, fake_mi355x_8):
+    sock = str(tmp_path / "r.sock")
+    plugin = AMDGPUPlugin(resource="gpu", paths=fake_mi355x_8.paths)
+    plugin.start()
+    for _ in range(3):
+        srv = NativePluginServer(plugin, sock)
+        srv.start()
+        ch = grpc.insecure_channel(f"unix://{sock}")
+        stub = dp.DevicePluginStub(ch)
+        assert stub.GetDevicePluginOptions(dp.Empty(), timeout=5) is not None
+        ch.close()
+        srv.stop()

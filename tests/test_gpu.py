@@ -173,3 +173,22 @@ def test_allocate_paths_are_real_devices(live_devices):
     render = [p for p in specs if "renderD" in p]
     assert render and os.path.exists(render[0])
     assert drm.dev_functional(render[0])
+
+
+def test_labeller_on_live_sysfs(live_devices):
+    """Label generation on the real MI355X: gfx950 values end-to-end."""
+    from k8s_device_plugin_amd.labeller import generate_labels
+    from k8s_device_plugin_amd.labeller.labels import LABEL_KINDS
+    from k8s_device_plugin_amd.topology import SysPaths
+
+    labels = generate_labels({k: True for k in LABEL_KINDS}, SysPaths("/"))
+    assert labels["amd.com/gpu.vram"] == "288G"
+    assert labels["amd.com/gpu.cu-count"] == "256"
+    assert labels["amd.com/gpu.simd-count"] == "1024"
+    assert labels["amd.com/gpu.device-id"] == "75a3"
+    assert labels["amd.com/gpu.family"] == "AI"
+    assert "MI355" in labels["amd.com/gpu.product-name"]
+    assert labels["amd.com/gpu.compute-memory-partition"] == "spx_nps1"
+    # firmware labels come from the raw-ioctl shim on a live box
+    fw_labels = [k for k in labels if k.startswith("beta.amd.com/gpu.firmware.")]
+    assert fw_labels, "expected firmware labels via drmctl ioctls"
