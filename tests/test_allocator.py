@@ -197,3 +197,54 @@ def test_cpx_allocate_all(fake_mi355x_cpx):
     ids = sorted(devices)
     out = policy.allocate(ids, [], 64)  # fast path: available == size
     assert len(out) == 64
+
+
+def test_mi308_like_4x8(tmp_path):
+    """4 GPUs x 8 partitions (the reference's mi308 topology shape):
+    8-partition requests pack one GPU; 12 spans exactly two."""
+    fs = build_mi355x_node(str(tmp_path / "m308"), n_gpus=4,
+                           partitions_per_gpu=8, compute_partition="CPX",
+                           memory_partition="NPS4")
+    policy, devices = make_policy(fs)
+    ids = sorted(devices)
+    assert len(ids) == 32
+    out = policy.allocate(ids, [], 8)
+    assert len({devices[i].dev_id for i in out}) == 1
+    out = policy.allocate(ids, [], 12)
+    assert len(out) == 12
+    assert len({devices[i].dev_id for i in out}) == 2
+
+
+def build_mixed_link_node(root):
+    """mi210-like: 8 whole GPUs, xGMI inside each NUMA quad, PCIe across
+    quads (reference topo-mi210-xgmi-pcie shape)."""
+    fs = FakeSysfs(root)
+    fs.add_cpu_node(0)
+    fs.add_cpu_node(1)
+    nodes = []
+    for i in range(8):
+        fs.add_physical_gpu(i, node_id=2 + i, numa_node=i // 4)
+        nodes.append(2 + i)
+    for a in range(8):
+        for b in range(a + 1, 8):
+            same_quad = (a // 4) == (b // 4)
+            fs.add_link(nodes[a], nodes[b],
+                        link_type=11 if same_quad else 2,
+                        weight=15 if same_quad else 40,
+                        bandwidth=153600 if same_quad else 64000)
+    return fs
+
+
+def test_mixed_links_prefer_xgmi(tmp_path):
+    """With xGMI quads and PCIe between them, a 3-GPU request from a mixed
+    window must stay inside one xGMI quad (reference
+    besteffort_policy_test.go:91-96 'same numa' case)."""
+    fs = build_mixed_link_node(str(tmp_path / "mixed"))
+    policy, devices = make_policy(fs)
+    ids = sorted(devices)
+    # window test3..test8 equivalent: GPUs 2..7 -> quad0: {2,3}, quad1: {4..7}
+    out = policy.allocate(ids[2:], [], 3)
+    assert set(out).issubset(set(ids[4:])), f"expected one-quad packing, got {out}"
+    # 4-GPU request: the whole xGMI quad
+    out = policy.allocate(ids[2:], [], 4)
+    assert set(out) == set(ids[4:])

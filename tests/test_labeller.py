@@ -158,3 +158,31 @@ def test_cli_labeller_oneshot(fake_mi355x_8, monkeypatch):
         assert fake.labels["amd.com/gpu.cu-count"] == "256"
     finally:
         fake.stop()
+
+
+def test_multi_value_counter_labels(tmp_path):
+    """Heterogeneous values produce counter labels in both namespaces and
+    no base label (reference createLabels semantics, main.go:87-108)."""
+    from k8s_device_plugin_amd.testing.fakesysfs import FakeSysfs
+
+    fs = FakeSysfs(str(tmp_path / "het"))
+    fs.add_cpu_node(0)
+    fs.add_physical_gpu(0, node_id=2, vram_bytes=309220868096)
+    fs.add_physical_gpu(1, node_id=3, vram_bytes=309220868096 // 2)
+    labels = generate_labels({"vram": True}, fs.paths)
+    assert "amd.com/gpu.vram" not in labels          # no single value
+    assert labels["amd.com/gpu.vram.288G"] == "1"
+    assert labels["amd.com/gpu.vram.144G"] == "1"
+    assert labels["beta.amd.com/gpu.vram.288G"] == "1"
+    assert "beta.amd.com/gpu.vram" not in labels
+
+
+def test_remove_sweeps_multi_value_counters():
+    labels = {
+        "amd.com/gpu.vram.288G": "1",
+        "amd.com/gpu.vram.144G": "1",
+        "beta.amd.com/gpu.vram.288G": "1",
+        "unrelated": "x",
+    }
+    remove_old_node_labels(labels)
+    assert labels == {"unrelated": "x"}
