@@ -89,10 +89,11 @@ __global__ void mfma_probe_kernel(float *d_out /* [2][256] */) {
 
 using f32x16 = __attribute__((ext_vector_type(16))) float;
 
-// Back-to-back v_mfma_f32_32x32x16_bf16 on independent accumulators:
-// 4 accumulators per wave cover the 32-cyc/SIMD issue + dependent latency,
-// 512-thread workgroups put 2 waves on each SIMD (guide: §Two waves per
-// SIMD — MFMA issue is fully paced per SIMD, partner waves interleave).
+// Back-to-back v_mfma_f32_32x32x16_bf16 on independent accumulators.
+// Measured sweep (gpurun_out/mfma_variants.txt): 2 accumulators at
+// 512 threads x 2 blocks/CU is fastest (2043 TF/s; 8 accumulators spill
+// VGPRs and collapse to 152 TF/s) — partner waves on each SIMD provide
+// the remaining issue cover (guide: §Two waves per SIMD).
 __global__ void __launch_bounds__(512, 2)
 mfma_peak_kernel(float *out, int iters) {
 #if defined(__gfx950__)
@@ -102,15 +103,13 @@ mfma_peak_kernel(float *out, int iters) {
         a[r] = (__bf16)(float)(((lane + r) % 5) - 2);
         b[r] = (__bf16)(float)(((lane * 3 + r) % 7) - 3);
     }
-    f32x16 acc0 = {}, acc1 = {}, acc2 = {}, acc3 = {};
+    f32x16 acc0 = {}, acc1 = {};
     for (int i = 0; i < iters; ++i) {
         acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc0, 0, 0, 0);
         acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc1, 0, 0, 0);
-        acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc2, 0, 0, 0);
-        acc3 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc3, 0, 0, 0);
     }
     float s = 0;
-    for (int r = 0; r < 16; ++r) s += acc0[r] + acc1[r] + acc2[r] + acc3[r];
+    for (int r = 0; r < 16; ++r) s += acc0[r] + acc1[r];
     if (s == -1.0f) out[blockIdx.x] = s;  // never true: defeats DCE only
 #else
     (void)out;
@@ -233,7 +232,7 @@ py::dict run_probe(int device, size_t hbm_bytes) {
     {
         float *d;
         HIP_CHECK(hipMalloc(&d, 4096 * sizeof(float)));
-        const int iters = 4096, blocks = props.multiProcessorCount * 2;
+        const int iters = 8192, blocks = props.multiProcessorCount * 2;
         hipLaunchKernelGGL(mfma_peak_kernel, dim3(blocks), dim3(512), 0, 0, d,
                            64);  // warmup
         HIP_CHECK(hipDeviceSynchronize());
@@ -247,8 +246,8 @@ py::dict run_probe(int device, size_t hbm_bytes) {
         HIP_CHECK(hipEventSynchronize(t1));
         float ms = 0;
         HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
-        // FLOPs: blocks * 8 waves * iters * 4 MFMA * 2*32*32*16
-        double flops = (double)blocks * 8 * iters * 4 * 2 * 32 * 32 * 16;
+        // FLOPs: blocks * 8 waves * iters * 2 MFMA * 2*32*32*16
+        double flops = (double)blocks * 8 * iters * 2 * 2 * 32 * 32 * 16;
         result["mfma_tflops"] = flops / (ms * 1e9);
         HIP_CHECK(hipEventDestroy(t0));
         HIP_CHECK(hipEventDestroy(t1));
