@@ -71,11 +71,20 @@ def main() -> int:
     result = {}
     harness = None
     if rank == 0:
-        harness = _Harness(args.gpus)
-        harness.start()
-        # warmup
-        for _ in range(args.warmup):
-            harness.step()
+        try:
+            harness = _Harness(args.gpus)
+            harness.start()
+            # warmup
+            for _ in range(args.warmup):
+                harness.step()
+        except Exception as e:
+            # fail the whole torchrun job instead of deadlocking peers at
+            # the timing barrier
+            import traceback
+
+            traceback.print_exc()
+            print(f"bench harness failed: {e}", file=sys.stderr, flush=True)
+            os._exit(1)
 
     barrier_sync()
     t0 = time.perf_counter()

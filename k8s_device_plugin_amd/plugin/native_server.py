@@ -62,8 +62,34 @@ class NativePluginServer:
         self._srv.start()
 
     def heartbeat(self) -> None:
-        """Recompute health and push the fresh list to every open stream."""
-        devs = self.plugin.refreshed_device_list()
+        """Recompute health and push the fresh list to every open stream.
+
+        Also re-walks sysfs: if the device set changed (hot-unplug,
+        partition-mode change), the Allocate fragments and allocator
+        tables are rebuilt — the reference only refreshes its cache when
+        the kubelet re-opens ListAndWatch (plugin.go:231)."""
+        p = self.plugin
+        from ..topology import discover_gpus
+
+        fresh = discover_gpus(p.paths, strict=False)
+        if set(fresh) != set(p.devices) or any(
+            fresh[i].render_d != p.devices[i].render_d for i in fresh
+        ):
+            log.info("device set changed (%d -> %d); rebuilding serving state",
+                     len(p.devices), len(fresh))
+            p.start()  # re-discover + re-init allocator
+            specs = {
+                d.id: _container_response_bytes(
+                    [f"/dev/dri/card{d.card}", f"/dev/dri/renderD{d.render_d}"]
+                )
+                for d in p.devices.values()
+            }
+            self._srv.set_device_specs(specs)
+            if not p.allocator_init_error and p.allocator.initialized:
+                groups, node_of_id, weights = p.allocator.export_state()
+                self._srv.set_allocator_state(groups, node_of_id, weights)
+
+        devs = p.refreshed_device_list()
         self._srv.push_list_update(
             dp.ListAndWatchResponse(devices=devs).SerializeToString()
         )

@@ -189,3 +189,60 @@ def test_cpx_preferred_parity(tmp_path):
         channel.close()
     finally:
         srv.stop()
+
+
+def test_heartbeat_rebuilds_on_device_change(tmp_path):
+    """Removing a GPU from sysfs between heartbeats must rebuild the
+    Allocate fragments and drop the device from the stream."""
+    import shutil
+    import queue
+    import threading
+
+    from k8s_device_plugin_amd.testing.fakesysfs import build_mi355x_node
+
+    fs = build_mi355x_node(str(tmp_path / "hp"), n_gpus=4)
+    sock = str(tmp_path / "hp.sock")
+    plugin = AMDGPUPlugin(resource="gpu", paths=fs.paths)
+    plugin.start()
+    srv = NativePluginServer(plugin, sock)
+    srv.start()
+    try:
+        channel = grpc.insecure_channel(f"unix://{sock}")
+        stub = dp.DevicePluginStub(channel)
+        call = stub.ListAndWatch(dp.Empty())
+        q = queue.Queue()
+
+        def reader():
+            try:
+                for r in call:
+                    q.put(r)
+            except Exception:
+                pass
+
+        threading.Thread(target=reader, daemon=True).start()
+        first = q.get(timeout=5)
+        assert len(first.devices) == 4
+        gone = sorted(plugin.devices)[3]
+
+        # remove one GPU's pci dir + kfd node
+        import os
+
+        shutil.rmtree(os.path.join(fs.paths.amdgpu_pci, gone))
+        node_id = plugin.devices[gone].node_id
+        shutil.rmtree(os.path.join(fs.paths.kfd_topology_nodes, str(node_id)))
+
+        srv.heartbeat()
+        second = q.get(timeout=5)
+        assert len(second.devices) == 3
+        assert gone not in {d.ID for d in second.devices}
+
+        # allocate for a surviving device still works
+        keep = sorted(plugin.devices)[0]
+        req = dp.AllocateRequest()
+        req.container_requests.add().devices_ids.append(keep)
+        resp = stub.Allocate(req, timeout=5)
+        assert len(resp.container_responses[0].devices) == 3
+        call.cancel()
+        channel.close()
+    finally:
+        srv.stop()
