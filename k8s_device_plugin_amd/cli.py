@@ -45,6 +45,11 @@ def device_plugin_main(argv=None) -> int:
     ap.add_argument("--server", default="native", choices=["native", "python"],
                     help="serving implementation (native C++/nghttp2 with "
                          "automatic python fallback, or python grpc)")
+    ap.add_argument("--cdi", action="store_true",
+                    help="write a CDI spec and return CDI device names from "
+                         "Allocate (containerd>=1.7 / CRI-O with CDI)")
+    ap.add_argument("--cdi-dir", default=None,
+                    help="CDI spec directory (default /var/run/cdi)")
     ap.add_argument("-v", "--verbose", action="count", default=0)
     args = ap.parse_args(argv)
     _setup_logging(args.verbose)
@@ -93,8 +98,17 @@ def device_plugin_main(argv=None) -> int:
         return 0
     log.info("advertising resources: %s", [f"amd.com/{r}" for r in resources])
 
+    if args.cdi:
+        from .plugin.cdi import CDI_SPEC_DIR, write_cdi_spec
+
+        spec_path = write_cdi_spec(
+            devices.values(), spec_dir=args.cdi_dir or CDI_SPEC_DIR
+        )
+        log.info("wrote CDI spec %s", spec_path)
+
     mgr = PluginManager(
-        lambda res: AMDGPUPlugin(resource=res, paths=paths),
+        lambda res: AMDGPUPlugin(resource=res, paths=paths,
+                                 cdi_enabled=args.cdi),
         device_plugin_path=args.kubelet_dir or dp.DEVICE_PLUGIN_PATH,
         server_impl=args.server,
     )

@@ -19,14 +19,17 @@ from .server import AMDGPUPlugin
 log = logging.getLogger(__name__)
 
 
-def _container_response_bytes(specs) -> bytes:
-    """Serialized ContainerAllocateResponse holding only `specs` — i.e. the
-    tagged `devices` field fragments, concatenable per protobuf rules."""
+def _container_response_bytes(specs, cdi_name=None) -> bytes:
+    """Serialized ContainerAllocateResponse holding only `specs` (and the
+    optional CDI device name) — tagged field fragments, concatenable per
+    protobuf rules."""
     car = dp.ContainerAllocateResponse()
     for host_path in specs:
         s = car.devices.add()
         s.host_path = s.container_path = host_path
         s.permissions = "rw"
+    if cdi_name:
+        car.cdi_devices.add().name = cdi_name
     return car.SerializeToString()
 
 
@@ -41,6 +44,13 @@ class NativePluginServer:
         self.socket_path = socket_path
         self._srv = mod.Server(socket_path)
 
+    def _cdi_name(self, device_id):
+        if not getattr(self.plugin, "cdi_enabled", False):
+            return None
+        from .cdi import cdi_device_name
+
+        return cdi_device_name(device_id)
+
     def start(self) -> None:
         p = self.plugin
         self._srv.set_options_response(
@@ -50,7 +60,8 @@ class NativePluginServer:
         specs: Dict[str, bytes] = {}
         for d in p.devices.values():
             specs[d.id] = _container_response_bytes(
-                [f"/dev/dri/card{d.card}", f"/dev/dri/renderD{d.render_d}"]
+                [f"/dev/dri/card{d.card}", f"/dev/dri/renderD{d.render_d}"],
+                cdi_name=self._cdi_name(d.id),
             )
         self._srv.set_device_specs(specs)
         self._srv.set_list_response(
@@ -80,7 +91,8 @@ class NativePluginServer:
             p.start()  # re-discover + re-init allocator
             specs = {
                 d.id: _container_response_bytes(
-                    [f"/dev/dri/card{d.card}", f"/dev/dri/renderD{d.render_d}"]
+                    [f"/dev/dri/card{d.card}", f"/dev/dri/renderD{d.render_d}"],
+                    cdi_name=self._cdi_name(d.id),
                 )
                 for d in p.devices.values()
             }

@@ -53,8 +53,10 @@ class AMDGPUPlugin:
         exporter_timeout: float = ms.QUERY_TIMEOUT_S,
         on_stream_lost: Optional[Callable[[], None]] = None,
         exit_on_stream_loss: bool = False,
+        cdi_enabled: bool = False,
     ):
         self.resource = resource
+        self.cdi_enabled = cdi_enabled
         self.paths = paths
         self.devices: Dict[str, GPUDevice] = {}
         self.allocator = allocator or BestEffortPolicy()
@@ -224,4 +226,8 @@ class AMDGPUPlugin:
                     spec = car.devices.add()
                     spec.host_path = spec.container_path = path
                     spec.permissions = "rw"
+                if self.cdi_enabled:
+                    from .cdi import cdi_device_name
+
+                    car.cdi_devices.add().name = cdi_device_name(dev_id)
         return response
