@@ -50,6 +50,8 @@ def device_plugin_main(argv=None) -> int:
                          "Allocate (containerd>=1.7 / CRI-O with CDI)")
     ap.add_argument("--cdi-dir", default=None,
                     help="CDI spec directory (default /var/run/cdi)")
+    ap.add_argument("--metrics-port", type=int, default=0,
+                    help="Prometheus metrics port (0 disables, the default)")
     ap.add_argument("-v", "--verbose", action="count", default=0)
     args = ap.parse_args(argv)
     _setup_logging(args.verbose)
@@ -114,6 +116,11 @@ def device_plugin_main(argv=None) -> int:
     )
     ticker = HeartbeatTicker(args.pulse)
     ticker.subscribe(mgr.heartbeat_all)
+
+    if args.metrics_port:
+        from .plugin.metrics import start_metrics_server
+
+        start_metrics_server(mgr, args.metrics_port)
 
     stop = threading.Event()
     for sig in (signal.SIGINT, signal.SIGTERM, signal.SIGQUIT):

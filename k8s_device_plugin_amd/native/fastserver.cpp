@@ -308,6 +308,19 @@ class Server {
         }
         wake();
     }
+    py::dict stats() {
+        py::dict d;
+        d["allocate_total"] = (uint64_t)n_allocate_.load();
+        d["preferred_allocation_total"] = (uint64_t)n_preferred_.load();
+        d["list_and_watch_streams_total"] = (uint64_t)n_listwatch_.load();
+        d["options_total"] = (uint64_t)n_options_.load();
+        d["prestart_total"] = (uint64_t)n_prestart_.load();
+        d["unknown_method_total"] = (uint64_t)n_unknown_.load();
+        d["list_pushes_total"] = (uint64_t)n_pushes_.load();
+        d["connections_total"] = (uint64_t)n_conns_.load();
+        return d;
+    }
+
     void set_allocator_state(
         const std::vector<std::pair<std::string, std::vector<int>>> &groups,
         const std::map<std::string, int> &node_of_id,
@@ -482,13 +495,17 @@ class Server {
         const std::string &p = st->path;
         std::string resp;
         if (p == "/v1beta1.DevicePlugin/GetDevicePluginOptions") {
+            n_options_.fetch_add(1, std::memory_order_relaxed);
             std::lock_guard<std::mutex> g(mu_);
             resp = options_;
         } else if (p == "/v1beta1.DevicePlugin/PreStartContainer") {
+            n_prestart_.fetch_add(1, std::memory_order_relaxed);
             resp = "";
         } else if (p == "/v1beta1.DevicePlugin/Allocate") {
+            n_allocate_.fetch_add(1, std::memory_order_relaxed);
             resp = handle_allocate(msg);
         } else if (p == "/v1beta1.DevicePlugin/GetPreferredAllocation") {
+            n_preferred_.fetch_add(1, std::memory_order_relaxed);
             std::string err;
             if (!handle_preferred(msg, resp, err)) {
                 st->grpc_status = "3";  // INVALID_ARGUMENT
@@ -498,6 +515,7 @@ class Server {
                 return;
             }
         } else if (p == "/v1beta1.DevicePlugin/ListAndWatch") {
+            n_listwatch_.fetch_add(1, std::memory_order_relaxed);
             st->is_listwatch = true;
             {
                 std::lock_guard<std::mutex> g(mu_);
@@ -507,6 +525,7 @@ class Server {
             submit_stream_response(conn, stream_id, st);
             return;
         } else {
+            n_unknown_.fetch_add(1, std::memory_order_relaxed);
             st->grpc_status = "12";  // UNIMPLEMENTED
             submit_unary(conn, stream_id, st, "", false);
             return;
@@ -629,6 +648,7 @@ class Server {
             nghttp2_settings_entry st[1] = {
                 {NGHTTP2_SETTINGS_MAX_CONCURRENT_STREAMS, 128}};
             ng.submit_settings(conn->session, NGHTTP2_FLAG_NONE, st, 1);
+            n_conns_.fetch_add(1, std::memory_order_relaxed);
             conns_.push_back(std::move(conn));
         }
     }
@@ -708,6 +728,7 @@ class Server {
                     bytes = list_bytes_;
                 }
                 if (do_push) {
+                    n_pushes_.fetch_add(1, std::memory_order_relaxed);
                     std::string framed = grpc_frame(bytes);
                     for (auto &c : conns_) {
                         for (auto &skv : c->streams) {
@@ -769,6 +790,10 @@ class Server {
     std::atomic<bool> running_{false};
     std::vector<std::unique_ptr<Conn>> conns_;
 
+    std::atomic<uint64_t> n_allocate_{0}, n_preferred_{0}, n_listwatch_{0},
+        n_options_{0}, n_prestart_{0}, n_unknown_{0}, n_pushes_{0},
+        n_conns_{0};
+
     std::mutex mu_;
     std::string options_, kfd_spec_, list_bytes_;
     std::unordered_map<std::string, std::string> dev_specs_;
@@ -790,6 +815,7 @@ PYBIND11_MODULE(_fastserver, m) {
         .def("set_list_response", &Server::set_list_response)
         .def("push_list_update", &Server::push_list_update)
         .def("set_allocator_state", &Server::set_allocator_state)
+        .def("stats", &Server::stats)
         .def("start", &Server::start,
              py::call_guard<py::gil_scoped_release>())
         .def("stop", &Server::stop, py::call_guard<py::gil_scoped_release>());
