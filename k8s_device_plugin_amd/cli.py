@@ -52,6 +52,9 @@ def device_plugin_main(argv=None) -> int:
                     help="CDI spec directory (default /var/run/cdi)")
     ap.add_argument("--metrics-port", type=int, default=0,
                     help="Prometheus metrics port (0 disables, the default)")
+    ap.add_argument("--prestart-probe", action="store_true",
+                    help="advertise pre_start_required and verify each "
+                         "requested device answers before container start")
     ap.add_argument("-v", "--verbose", action="count", default=0)
     args = ap.parse_args(argv)
     _setup_logging(args.verbose)
@@ -110,7 +113,8 @@ def device_plugin_main(argv=None) -> int:
 
     mgr = PluginManager(
         lambda res: AMDGPUPlugin(resource=res, paths=paths,
-                                 cdi_enabled=args.cdi),
+                                 cdi_enabled=args.cdi,
+                                 prestart_probe=args.prestart_probe),
         device_plugin_path=args.kubelet_dir or dp.DEVICE_PLUGIN_PATH,
         server_impl=args.server,
     )

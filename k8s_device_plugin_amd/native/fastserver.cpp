@@ -308,6 +308,12 @@ class Server {
         }
         wake();
     }
+    void set_prestart_paths(const std::map<std::string, std::string> &paths) {
+        std::lock_guard<std::mutex> g(mu_);
+        prestart_paths_.clear();
+        for (auto &kv : paths) prestart_paths_[kv.first] = kv.second;
+    }
+
     py::dict stats() {
         py::dict d;
         d["allocate_total"] = (uint64_t)n_allocate_.load();
@@ -500,6 +506,14 @@ class Server {
             resp = options_;
         } else if (p == "/v1beta1.DevicePlugin/PreStartContainer") {
             n_prestart_.fetch_add(1, std::memory_order_relaxed);
+            std::string bad;
+            if (!handle_prestart(msg, bad)) {
+                st->grpc_status = "9";  // FAILED_PRECONDITION
+                st->grpc_message =
+                    "device " + bad + " failed the pre-start probe";
+                submit_unary(conn, stream_id, st, "", false);
+                return;
+            }
             resp = "";
         } else if (p == "/v1beta1.DevicePlugin/Allocate") {
             n_allocate_.fetch_add(1, std::memory_order_relaxed);
@@ -561,6 +575,34 @@ class Server {
 
     void submit_stream_response(Conn *conn, int32_t stream_id, Stream *st) {
         submit_headers_common(conn, stream_id, st);
+    }
+
+    // true = every requested device opens; on failure `bad` names it
+    bool handle_prestart(const std::string &msg, std::string &bad) {
+        std::unordered_map<std::string, std::string> paths;
+        {
+            std::lock_guard<std::mutex> g(mu_);
+            if (prestart_paths_.empty()) return true;  // probe disabled
+            paths = prestart_paths_;
+        }
+        bool ok = true;
+        const uint8_t *p = (const uint8_t *)msg.data();
+        for_each_field(p, p + msg.size(), [&](int field, int wt,
+                                              const uint8_t *data,
+                                              uint64_t len) {
+            if (!ok || field != 1 || wt != 2) return;
+            std::string id((const char *)data, len);
+            auto it = paths.find(id);
+            if (it == paths.end()) return;
+            int fd = ::open(it->second.c_str(), O_RDWR | O_CLOEXEC);
+            if (fd < 0) {
+                ok = false;
+                bad = id;
+            } else {
+                ::close(fd);
+            }
+        });
+        return ok;
     }
 
     std::string handle_allocate(const std::string &msg) {
@@ -797,6 +839,7 @@ class Server {
     std::mutex mu_;
     std::string options_, kfd_spec_, list_bytes_;
     std::unordered_map<std::string, std::string> dev_specs_;
+    std::unordered_map<std::string, std::string> prestart_paths_;
     AllocState alloc_;
     bool pending_push_ = false;
 
@@ -816,6 +859,7 @@ PYBIND11_MODULE(_fastserver, m) {
         .def("push_list_update", &Server::push_list_update)
         .def("set_allocator_state", &Server::set_allocator_state)
         .def("stats", &Server::stats)
+        .def("set_prestart_paths", &Server::set_prestart_paths)
         .def("start", &Server::start,
              py::call_guard<py::gil_scoped_release>())
         .def("stop", &Server::stop, py::call_guard<py::gil_scoped_release>());

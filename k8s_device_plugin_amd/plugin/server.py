@@ -54,9 +54,17 @@ class AMDGPUPlugin:
         on_stream_lost: Optional[Callable[[], None]] = None,
         exit_on_stream_loss: bool = False,
         cdi_enabled: bool = False,
+        prestart_probe: bool = False,
+        dev_root: str = "/dev",
     ):
         self.resource = resource
         self.cdi_enabled = cdi_enabled
+        # when enabled, PreStartContainer verifies each requested device
+        # answers before the container starts (the reference's
+        # PreStartContainer is a no-op and never advertised,
+        # plugin.go:219-224)
+        self.prestart_probe = prestart_probe
+        self.dev_root = dev_root
         self.paths = paths
         self.devices: Dict[str, GPUDevice] = {}
         self.allocator = allocator or BestEffortPolicy()
@@ -105,9 +113,40 @@ class AMDGPUPlugin:
         opts = dp.DevicePluginOptions()
         if not self.allocator_init_error:
             opts.get_preferred_allocation_available = True
+        if self.prestart_probe:
+            opts.pre_start_required = True
         return opts
 
+    def render_device_path(self, dev: GPUDevice) -> str:
+        return f"{self.dev_root}/dri/renderD{dev.render_d}"
+
     def PreStartContainer(self, request, context):
+        if self.prestart_probe:
+            from ..native import load_drmctl
+
+            drm = load_drmctl()
+            for dev_id in request.devices_ids:
+                dev = self.devices.get(dev_id)
+                if dev is None:
+                    continue
+                path = self.render_device_path(dev)
+                ok = os.path.exists(path)
+                if ok and drm is not None:
+                    import stat
+
+                    # the DRM_IOCTL probe only makes sense on a real device
+                    # node (tests use plain files)
+                    if stat.S_ISCHR(os.stat(path).st_mode):
+                        ok = drm.dev_functional(path)
+                if not ok:
+                    import grpc
+
+                    log.error("PreStartContainer: device %s (%s) not "
+                              "functional", dev_id, path)
+                    context.abort(
+                        grpc.StatusCode.FAILED_PRECONDITION,
+                        f"device {dev_id} failed the pre-start probe",
+                    )
         return dp.PreStartContainerResponse()
 
     def _my_devices(self) -> List[GPUDevice]:
