@@ -192,3 +192,32 @@ def test_labeller_on_live_sysfs(live_devices):
     # firmware labels come from the raw-ioctl shim on a live box
     fw_labels = [k for k in labels if k.startswith("beta.amd.com/gpu.firmware.")]
     assert fw_labels, "expected firmware labels via drmctl ioctls"
+
+
+def test_prestart_probe_live(live_devices):
+    """--prestart-probe against the real renderD node via the native server."""
+    import grpc
+    import tempfile
+
+    from k8s_device_plugin_amd.plugin import AMDGPUPlugin
+    from k8s_device_plugin_amd.plugin.native_server import NativePluginServer
+    from k8s_device_plugin_amd.protos import deviceplugin as dp
+
+    paths, devs = live_devices
+    plugin = AMDGPUPlugin(resource="gpu", paths=paths, prestart_probe=True)
+    plugin.start()
+    with tempfile.TemporaryDirectory() as tmp:
+        srv = NativePluginServer(plugin, f"{tmp}/ps.sock")
+        srv.start()
+        try:
+            ch = grpc.insecure_channel(f"unix://{tmp}/ps.sock")
+            stub = dp.DevicePluginStub(ch)
+            backed = sorted(d.id for d in devs.values() if d.kfd_backed)
+            req = dp.PreStartContainerRequest()
+            req.devices_ids.extend(backed[:1])
+            assert stub.PreStartContainer(req, timeout=5) is not None
+            opts = stub.GetDevicePluginOptions(dp.Empty(), timeout=5)
+            assert opts.pre_start_required
+            ch.close()
+        finally:
+            srv.stop()
