@@ -163,11 +163,7 @@ class _Harness:
         from k8s_device_plugin_amd.plugin import AMDGPUPlugin, PluginManager
         from k8s_device_plugin_amd.protos import deviceplugin as dp
         from k8s_device_plugin_amd.testing.stub_kubelet import StubKubelet
-        from k8s_device_plugin_amd.topology import (
-            SysPaths,
-            discover_gpus,
-            simple_health_check,
-        )
+        from k8s_device_plugin_amd.topology import SysPaths, simple_health_check
 
         self._tmp = tempfile.TemporaryDirectory(prefix="amdxdp-bench-")
         root = self._tmp.name

@@ -18,7 +18,7 @@ The pair weights themselves are hive-aware (weights.py), which on an
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, Iterable, List, Optional, Sequence
 
 from ..topology.discovery import GPUDevice

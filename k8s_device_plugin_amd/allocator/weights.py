@@ -15,8 +15,7 @@ reference on topologies without hive information.
 
 from __future__ import annotations
 
-from dataclasses import dataclass
-from typing import Dict, Iterable, List, Optional
+from typing import Dict, Iterable
 
 from ..topology.discovery import GPUDevice
 from ..topology.kfd import KFDLink, KFDTopology

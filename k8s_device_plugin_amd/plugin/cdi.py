@@ -14,7 +14,7 @@ from __future__ import annotations
 import json
 import os
 import tempfile
-from typing import Dict, Iterable
+from typing import Iterable
 
 from ..topology.discovery import GPUDevice
 

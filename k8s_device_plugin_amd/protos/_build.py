@@ -11,7 +11,7 @@ internal/pkg/exporter/metricssvc/metricssvc.pb.go).
 
 from __future__ import annotations
 
-from typing import Dict, List, Sequence, Tuple
+from typing import Dict, Sequence, Tuple
 
 from google.protobuf import descriptor_pb2, descriptor_pool, message_factory
 

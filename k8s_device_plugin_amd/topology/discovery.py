@@ -18,7 +18,7 @@ import glob
 import os
 import re
 from dataclasses import dataclass, asdict
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 from .kfd import KFDTopology
 from .sysfs import SysPaths, read_stripped

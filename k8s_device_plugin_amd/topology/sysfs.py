@@ -10,7 +10,7 @@ tests and benchmarks with one argument.
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 
 @dataclass(frozen=True)
