@@ -3,7 +3,7 @@ IMAGE_DP ?= rocm/k8s-device-plugin-mi355x
 IMAGE_NL ?= rocm/k8s-node-labeller-mi355x
 TAG ?= $(shell git describe --always --dirty 2>/dev/null || echo dev)
 
-.PHONY: all native test test-gpu bench images dp-image labeller-image helm clean
+.PHONY: all native test test-gpu bench images dp-image labeller-image ubi-dp-image ubi-labeller-image helm clean
 
 all: native
 
@@ -19,13 +19,19 @@ test-gpu:
 bench:
 	python3 bench.py --gpus 1 --steps 500 --warmup 50
 
-images: dp-image labeller-image
+images: dp-image labeller-image ubi-dp-image ubi-labeller-image
 
 dp-image:
 	docker build -f deploy/Dockerfile -t $(IMAGE_DP):$(TAG) .
 
 labeller-image:
 	docker build -f deploy/labeller.Dockerfile -t $(IMAGE_NL):$(TAG) .
+
+ubi-dp-image:
+	docker build -f deploy/ubi-dp.Dockerfile -t $(IMAGE_DP):$(TAG)-ubi .
+
+ubi-labeller-image:
+	docker build -f deploy/ubi-labeller.Dockerfile -t $(IMAGE_NL):$(TAG)-ubi .
 
 helm:
 	helm lint deploy/helm/amd-gpu
