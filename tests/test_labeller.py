@@ -186,3 +186,35 @@ def test_remove_sweeps_multi_value_counters():
     }
     remove_old_node_labels(labels)
     assert labels == {"unrelated": "x"}
+
+
+def test_watch_reconnects_after_stream_end(fake_mi355x_8):
+    """The watch stream ends (API server timeout); the controller must
+    reconnect and keep reconciling (informer relist behavior)."""
+    import time
+
+    fake = FakeK8s(node_name="n2").start()
+    try:
+        labels = generate_labels({"vram": True}, fake_mi355x_8.paths)
+        client = K8sClient(base_url=fake.base_url)
+        ctl = NodeLabelController(client, "n2", labels)
+        ctl.run(block=False)
+        # first event on the first stream
+        fake.push_event("ADDED")
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline and "amd.com/gpu.vram" not in fake.labels:
+            time.sleep(0.05)
+        assert fake.labels.get("amd.com/gpu.vram") == "288G"
+
+        # the fake stream times out after 5s of no events; push another
+        # event afterwards — only a reconnected watcher can see it
+        fake.labels.clear()
+        time.sleep(6.0)
+        fake.push_event("ADDED")
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline and "amd.com/gpu.vram" not in fake.labels:
+            time.sleep(0.05)
+        assert fake.labels.get("amd.com/gpu.vram") == "288G", "watch did not reconnect"
+        ctl.stop()
+    finally:
+        fake.stop()
