@@ -11,6 +11,17 @@ def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: test requires a real AMD GPU (run on an MI355X box)"
     )
+    # build the CPU-buildable native extensions if missing so a fresh
+    # checkout passes without a separate build step (g++ only; the HIP
+    # probe needs hipcc and is attempted best-effort)
+    from k8s_device_plugin_amd.native import build as nb
+
+    nb.build_drmctl()
+    nb.build_fastserver()
+    try:
+        nb.build_healthprobe()
+    except Exception:
+        pass  # no hipcc on this machine; gpu-marked tests need it anyway
 
 
 @pytest.fixture
