@@ -184,7 +184,8 @@ class PluginManager:
         self._watch_thread.start()
 
     def heartbeat_all(self) -> None:
-        for inst in self.plugins.values():
+        # snapshot: the ticker thread may race a start_resource dict insert
+        for inst in list(self.plugins.values()):
             if inst.native:
                 inst.server.heartbeat()  # recompute + push to all streams
             else:
