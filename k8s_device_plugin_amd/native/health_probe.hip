@@ -334,6 +334,55 @@ int device_count() {
     return n;
 }
 
+// Measured GPU->GPU copy bandwidth: on an 8*MI355X hive this runs over one
+// xGMI point-to-point link (~153 GB/s class); PCIe-bridged pairs land far
+// lower — the hardware truth behind the allocator's hive packing
+// (docs/resource-allocation.md).
+py::dict p2p_bandwidth(int src, int dst, size_t bytes) {
+    py::dict out;
+    out["src"] = src;
+    out["dst"] = dst;
+
+    int can = 0;
+    HIP_CHECK(hipDeviceCanAccessPeer(&can, dst, src));
+    out["peer_access"] = (bool)can;
+
+    HIP_CHECK(hipSetDevice(src));
+    void *src_buf;
+    HIP_CHECK(hipMalloc(&src_buf, bytes));
+    HIP_CHECK(hipMemset(src_buf, 0x5A, bytes));
+    HIP_CHECK(hipSetDevice(dst));
+    if (can) {
+        hipError_t e = hipDeviceEnablePeerAccess(src, 0);
+        if (e != hipSuccess && e != hipErrorPeerAccessAlreadyEnabled)
+            throw std::runtime_error(hipGetErrorString(e));
+    }
+    void *dst_buf;
+    HIP_CHECK(hipMalloc(&dst_buf, bytes));
+
+    const int iters = 10;
+    HIP_CHECK(hipMemcpyPeerAsync(dst_buf, dst, src_buf, src, bytes, 0));
+    HIP_CHECK(hipDeviceSynchronize());
+    hipEvent_t t0, t1;
+    HIP_CHECK(hipEventCreate(&t0));
+    HIP_CHECK(hipEventCreate(&t1));
+    HIP_CHECK(hipEventRecord(t0));
+    for (int i = 0; i < iters; ++i)
+        HIP_CHECK(hipMemcpyPeerAsync(dst_buf, dst, src_buf, src, bytes, 0));
+    HIP_CHECK(hipEventRecord(t1));
+    HIP_CHECK(hipEventSynchronize(t1));
+    float ms = 0;
+    HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+    out["gbps"] = ((double)bytes * iters) / (ms * 1e6);
+
+    HIP_CHECK(hipEventDestroy(t0));
+    HIP_CHECK(hipEventDestroy(t1));
+    HIP_CHECK(hipFree(dst_buf));
+    HIP_CHECK(hipSetDevice(src));
+    HIP_CHECK(hipFree(src_buf));
+    return out;
+}
+
 } // namespace
 
 PYBIND11_MODULE(_healthprobe, m) {
@@ -341,4 +390,6 @@ PYBIND11_MODULE(_healthprobe, m) {
     m.def("run_probe", &run_probe, py::arg("device") = 0,
           py::arg("hbm_bytes") = (size_t)1 << 30);
     m.def("device_count", &device_count);
+    m.def("p2p_bandwidth", &p2p_bandwidth, py::arg("src") = 0,
+          py::arg("dst") = 1, py::arg("bytes") = (size_t)1 << 30);
 }
