@@ -240,3 +240,51 @@ def test_two_streams_both_get_heartbeats(harness):
     assert len(r1.devices) == len(r2.devices) == 8
     call1.cancel()
     call2.cancel()
+
+
+def test_register_fails_gracefully_without_kubelet(tmp_path, fake_mi355x_8):
+    """No kubelet.sock: registration retries then returns False; the plugin
+    keeps serving and registers when the kubelet appears."""
+    dp_dir = str(tmp_path / "dp")
+    import os
+
+    os.makedirs(dp_dir, exist_ok=True)
+    mgr = PluginManager(
+        lambda r: AMDGPUPlugin(resource=r, paths=fake_mi355x_8.paths),
+        device_plugin_path=dp_dir, watch_interval=0.1,
+    )
+    try:
+        mgr.run(["gpu"])  # registration fails (no kubelet yet) but serving is up
+        kubelet = StubKubelet(dp_dir).start()
+        try:
+            reg = kubelet.wait_for_registration(timeout=10)
+            assert reg.resource_name == "amd.com/gpu"
+        finally:
+            kubelet.stop()
+    finally:
+        mgr.stop()
+
+
+def test_native_unavailable_falls_back(tmp_path, fake_mi355x_8, monkeypatch):
+    """If the fast server can't load, the manager serves via python grpc."""
+    import k8s_device_plugin_amd.plugin.native_server as ns
+
+    def boom(*a, **k):
+        raise RuntimeError("simulated native load failure")
+
+    monkeypatch.setattr(ns, "NativePluginServer", boom)
+    dp_dir = str(tmp_path / "dp")
+    kubelet = StubKubelet(dp_dir).start()
+    mgr = PluginManager(
+        lambda r: AMDGPUPlugin(resource=r, paths=fake_mi355x_8.paths),
+        device_plugin_path=dp_dir, server_impl="native",
+    )
+    try:
+        mgr.run(["gpu"])
+        assert not mgr.plugins["gpu"].native
+        reg = kubelet.wait_for_registration()
+        stub = kubelet.connect(reg.endpoint)
+        assert stub.GetDevicePluginOptions(dp.Empty(), timeout=5) is not None
+    finally:
+        mgr.stop()
+        kubelet.stop()
