@@ -106,6 +106,7 @@ def main() -> int:
             statistics.median(harness.pref_lat_us) if harness.pref_lat_us else None
         )
         advertised = harness.advertised
+        native_client_p50 = harness.native_client_p50_us()
         harness.stop()
 
         value = args.gpus * args.steps / elapsed  # device-grants/s, whole job
@@ -132,6 +133,7 @@ def main() -> int:
                 "advertised_gpus": advertised,
                 "present_gpus": present_gpus,
                 "allocate_p50_us": round(alloc_p50_us, 1),
+                "native_client_allocate_p50_us": native_client_p50,
                 "preferred_alloc_p50_us": (
                     round(pref_p50_us, 1) if pref_p50_us is not None else None
                 ),
@@ -243,6 +245,40 @@ class _Harness:
         assert len(specs) == 1 + 2 * self.n, (
             f"expected /dev/kfd + 2 nodes per device, got {len(specs)}"
         )
+
+    def native_client_p50_us(self):
+        """Allocate p50 measured with the C/nghttp2 bench client — the
+        latency a compiled kubelet sees, without Python-client overhead."""
+        import subprocess
+
+        exe = os.path.join(
+            os.path.dirname(os.path.abspath(__file__)),
+            "k8s_device_plugin_amd", "native", "benchclient",
+        )
+        src = exe + ".cpp"
+        if not os.path.exists(exe) and os.path.exists(src):
+            try:
+                subprocess.run(
+                    ["g++", "-O2", "-std=c++17", src, "-o", exe, "-ldl"],
+                    check=True, capture_output=True, timeout=120,
+                )
+            except Exception:
+                return None
+        if not os.path.exists(exe) or not getattr(self, "server_impl", "") == "native":
+            return None
+        sock = self._mgr.plugins["gpu"].socket_path
+        try:
+            out = subprocess.run(
+                [exe, sock, self.device_ids[0], "2000"],
+                capture_output=True, text=True, timeout=120,
+            )
+            if out.returncode != 0:
+                return None
+            return json.loads(out.stdout.strip().splitlines()[-1])[
+                "allocate_p50_us"
+            ]
+        except Exception:
+            return None
 
     def stop(self) -> None:
         if self._stream_call is not None:

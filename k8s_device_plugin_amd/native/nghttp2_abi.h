@@ -137,6 +137,12 @@ struct NgHttp2 {
                           size_t);
     int (*session_resume_data)(nghttp2_session *, int32_t);
     int (*submit_rst_stream)(nghttp2_session *, uint8_t, int32_t, uint32_t);
+    // client side (used by the native bench client)
+    int (*session_client_new)(nghttp2_session **,
+                              const nghttp2_session_callbacks *, void *);
+    int32_t (*submit_request)(nghttp2_session *, const void *priority_spec,
+                              const nghttp2_nv *, size_t,
+                              const nghttp2_data_provider *, void *);
 
     static NgHttp2 &get() {
         static NgHttp2 inst;
@@ -179,6 +185,8 @@ struct NgHttp2 {
         LOAD(submit_trailer, "nghttp2_submit_trailer");
         LOAD(session_resume_data, "nghttp2_session_resume_data");
         LOAD(submit_rst_stream, "nghttp2_submit_rst_stream");
+        LOAD(session_client_new, "nghttp2_session_client_new");
+        LOAD(submit_request, "nghttp2_submit_request");
 #undef LOAD
     }
 };
