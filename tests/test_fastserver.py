@@ -246,3 +246,36 @@ def test_heartbeat_rebuilds_on_device_change(tmp_path):
         channel.close()
     finally:
         srv.stop()
+
+
+def test_stream_and_unary_same_connection(native):
+    """kubelet multiplexes ListAndWatch and Allocate on one connection;
+    both must progress concurrently on a single channel."""
+    import queue
+    import threading
+
+    plugin, srv, stub = native
+    call = stub.ListAndWatch(dp.Empty())
+    q = queue.Queue()
+
+    def reader():
+        try:
+            for r in call:
+                q.put(r)
+        except Exception:
+            pass
+
+    threading.Thread(target=reader, daemon=True).start()
+    q.get(timeout=5)
+
+    ids = sorted(plugin.devices)
+    req = dp.AllocateRequest()
+    req.container_requests.add().devices_ids.append(ids[0])
+    for _ in range(20):
+        resp = stub.Allocate(req, timeout=5)
+        assert len(resp.container_responses[0].devices) == 3
+    srv.heartbeat()
+    assert len(q.get(timeout=5).devices) == 8
+    resp = stub.Allocate(req, timeout=5)
+    assert len(resp.container_responses[0].devices) == 3
+    call.cancel()
