@@ -55,6 +55,10 @@ def device_plugin_main(argv=None) -> int:
     ap.add_argument("--prestart-probe", action="store_true",
                     help="advertise pre_start_required and verify each "
                          "requested device answers before container start")
+    ap.add_argument("--exit-on-stream-loss", action="store_true",
+                    help="exit(1) when a ListAndWatch stream breaks so the "
+                         "DaemonSet restarts the pod (the ROCm plugin's "
+                         "behavior); default is in-process re-registration")
     ap.add_argument("-v", "--verbose", action="count", default=0)
     args = ap.parse_args(argv)
     _setup_logging(args.verbose)
@@ -114,7 +118,8 @@ def device_plugin_main(argv=None) -> int:
     mgr = PluginManager(
         lambda res: AMDGPUPlugin(resource=res, paths=paths,
                                  cdi_enabled=args.cdi,
-                                 prestart_probe=args.prestart_probe),
+                                 prestart_probe=args.prestart_probe,
+                                 exit_on_stream_loss=args.exit_on_stream_loss),
         device_plugin_path=args.kubelet_dir or dp.DEVICE_PLUGIN_PATH,
         server_impl=args.server,
     )

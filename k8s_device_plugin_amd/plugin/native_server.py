@@ -104,6 +104,14 @@ class NativePluginServer:
             if not p.allocator_init_error and p.allocator.initialized:
                 groups, node_of_id, weights = p.allocator.export_state()
                 self._srv.set_allocator_state(groups, node_of_id, weights)
+            if p.cdi_enabled:
+                # keep the CDI spec in step with the device set
+                try:
+                    from .cdi import write_cdi_spec
+
+                    write_cdi_spec(p.devices.values())
+                except OSError as e:
+                    log.warning("CDI spec refresh failed: %s", e)
 
         devs = p.refreshed_device_list()
         self._srv.push_list_update(
