@@ -118,6 +118,7 @@ def device_plugin_main(argv=None) -> int:
     mgr = PluginManager(
         lambda res: AMDGPUPlugin(resource=res, paths=paths,
                                  cdi_enabled=args.cdi,
+                                 cdi_spec_dir=args.cdi_dir,
                                  prestart_probe=args.prestart_probe,
                                  exit_on_stream_loss=args.exit_on_stream_loss),
         device_plugin_path=args.kubelet_dir or dp.DEVICE_PLUGIN_PATH,

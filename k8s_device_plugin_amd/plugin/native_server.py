@@ -107,9 +107,12 @@ class NativePluginServer:
             if p.cdi_enabled:
                 # keep the CDI spec in step with the device set
                 try:
-                    from .cdi import write_cdi_spec
+                    from .cdi import CDI_SPEC_DIR, write_cdi_spec
 
-                    write_cdi_spec(p.devices.values())
+                    write_cdi_spec(
+                        p.devices.values(),
+                        spec_dir=p.cdi_spec_dir or CDI_SPEC_DIR,
+                    )
                 except OSError as e:
                     log.warning("CDI spec refresh failed: %s", e)
 

@@ -54,11 +54,13 @@ class AMDGPUPlugin:
         on_stream_lost: Optional[Callable[[], None]] = None,
         exit_on_stream_loss: bool = False,
         cdi_enabled: bool = False,
+        cdi_spec_dir: Optional[str] = None,
         prestart_probe: bool = False,
         dev_root: str = "/dev",
     ):
         self.resource = resource
         self.cdi_enabled = cdi_enabled
+        self.cdi_spec_dir = cdi_spec_dir
         # when enabled, PreStartContainer verifies each requested device
         # answers before the container starts (the reference's
         # PreStartContainer is a no-op and never advertised,

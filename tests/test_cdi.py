@@ -76,3 +76,31 @@ def test_allocate_with_cdi_native(tmp_path, fake_mi355x_8):
         ch.close()
     finally:
         srv.stop()
+
+
+def test_cdi_spec_refreshed_on_device_change(tmp_path):
+    """Removing a GPU between heartbeats must rewrite the CDI spec."""
+    import os
+    import shutil
+
+    from k8s_device_plugin_amd.testing.fakesysfs import build_mi355x_node
+
+    fs = build_mi355x_node(str(tmp_path / "r"), n_gpus=4)
+    cdi_dir = str(tmp_path / "cdi")
+    plugin = AMDGPUPlugin(resource="gpu", paths=fs.paths, cdi_enabled=True,
+                          cdi_spec_dir=cdi_dir)
+    plugin.start()
+    write_cdi_spec(plugin.devices.values(), spec_dir=cdi_dir)
+    srv = NativePluginServer(plugin, str(tmp_path / "c.sock"))
+    srv.start()
+    try:
+        gone = sorted(plugin.devices)[3]
+        node_id = plugin.devices[gone].node_id
+        shutil.rmtree(os.path.join(fs.paths.amdgpu_pci, gone))
+        shutil.rmtree(os.path.join(fs.paths.kfd_topology_nodes, str(node_id)))
+        srv.heartbeat()
+        spec = json.load(open(os.path.join(cdi_dir, "amd.com-gpu.json")))
+        assert len(spec["devices"]) == 3
+        assert gone not in [d["name"] for d in spec["devices"]]
+    finally:
+        srv.stop()
