@@ -104,6 +104,10 @@ class NativePluginServer:
             if not p.allocator_init_error and p.allocator.initialized:
                 groups, node_of_id, weights = p.allocator.export_state()
                 self._srv.set_allocator_state(groups, node_of_id, weights)
+            if p.prestart_probe:
+                self._srv.set_prestart_paths(
+                    {d.id: p.render_device_path(d) for d in p.devices.values()}
+                )
             if p.cdi_enabled:
                 # keep the CDI spec in step with the device set
                 try:
