@@ -51,10 +51,13 @@ def main() -> int:
         dist = dist_mod
         dist.init_process_group(backend="gloo")
 
-    cuda = torch.cuda.is_available()
+    present_gpus = torch.cuda.device_count() if torch.cuda.is_available() else 0
+    # a rank only touches a GPU that actually exists (a world size larger
+    # than the visible device count happens when the CPU-path bench is
+    # launched on a partially-visible box)
+    cuda = torch.cuda.is_available() and local_rank < present_gpus
 
     # ---- advertised-vs-present: every rank touches its own GPU ----
-    present_gpus = torch.cuda.device_count() if cuda else 0
     if cuda:
         torch.cuda.set_device(local_rank)
         x = torch.ones(1024, 1024, device=f"cuda:{local_rank}")
