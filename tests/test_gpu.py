@@ -240,3 +240,68 @@ def test_xgmi_p2p_bandwidth():
     # rate; assert a conservative floor and report the measured value
     assert r["gbps"] > 15, f"p2p bandwidth {r['gbps']:.0f} GB/s: link broken?"
     print(f"p2p 0->1: {r['gbps']:.0f} GB/s")
+
+
+def test_full_stack_daemon_live(tmp_path):
+    """The real CLI daemon on live sysfs with pulse + CDI + metrics:
+    registration, stream, allocate, health refresh, CDI spec, metrics."""
+    _require_gpu()
+    import json as _json
+    import signal
+    import socket as _socket
+    import subprocess
+    import urllib.request
+
+    from k8s_device_plugin_amd.testing.stub_kubelet import StubKubelet
+    from k8s_device_plugin_amd.protos import deviceplugin as dp
+
+    with _socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        mport = s.getsockname()[1]
+
+    dp_dir = str(tmp_path / "dp")
+    cdi_dir = str(tmp_path / "cdi")
+    kubelet = StubKubelet(dp_dir).start()
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "k8s_device_plugin_amd.cli",
+         "--pulse", "1", "--kubelet-dir", dp_dir,
+         "--cdi", "--cdi-dir", cdi_dir,
+         "--metrics-port", str(mport)],
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+    )
+    try:
+        reg = kubelet.wait_for_registration(timeout=60)
+        assert reg.resource_name == "amd.com/gpu"
+        stub = kubelet.connect(reg.endpoint)
+        call = stub.ListAndWatch(dp.Empty())
+        it = iter(call)
+        first = next(it)
+        healthy = [d.ID for d in first.devices if d.health == "Healthy"]
+        assert healthy
+        second = next(it)  # pulse=1 refresh
+        assert {d.ID for d in second.devices} == {d.ID for d in first.devices}
+
+        req = dp.AllocateRequest()
+        req.container_requests.add().devices_ids.append(healthy[0])
+        resp = stub.Allocate(req, timeout=10)
+        car = resp.container_responses[0]
+        assert any("/dev/kfd" == s.host_path for s in car.devices)
+        assert [c.name for c in car.cdi_devices] == [
+            f"amd.com/gpu={healthy[0]}"
+        ]
+
+        spec = _json.load(open(os.path.join(cdi_dir, "amd.com-gpu.json")))
+        assert any(d["name"] == healthy[0] for d in spec["devices"])
+
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{mport}/metrics", timeout=10
+        ).read().decode()
+        assert "amdgpu_dp_allocate_total" in body
+        call.cancel()
+        proc.send_signal(signal.SIGTERM)
+        assert proc.wait(timeout=20) == 0
+    finally:
+        if proc.poll() is None:
+            proc.kill()
+            proc.wait()
+        kubelet.stop()
