@@ -104,7 +104,9 @@ def main() -> int:
         elapsed = float(t[0])
 
     if rank == 0:
-        alloc_p50_us = statistics.median(harness.alloc_lat_us)
+        lat_sorted = sorted(harness.alloc_lat_us)
+        alloc_p50_us = statistics.median(lat_sorted)
+        alloc_p99_us = lat_sorted[int(len(lat_sorted) * 0.99)]
         pref_p50_us = (
             statistics.median(harness.pref_lat_us) if harness.pref_lat_us else None
         )
@@ -136,6 +138,7 @@ def main() -> int:
                 "advertised_gpus": advertised,
                 "present_gpus": present_gpus,
                 "allocate_p50_us": round(alloc_p50_us, 1),
+                "allocate_p99_us": round(alloc_p99_us, 1),
                 "native_client_allocate_p50_us": native_client_p50,
                 "preferred_alloc_p50_us": (
                     round(pref_p50_us, 1) if pref_p50_us is not None else None
