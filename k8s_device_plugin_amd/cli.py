@@ -55,6 +55,9 @@ def device_plugin_main(argv=None) -> int:
     ap.add_argument("--prestart-probe", action="store_true",
                     help="advertise pre_start_required and verify each "
                          "requested device answers before container start")
+    ap.add_argument("--dump", action="store_true",
+                    help="print discovered devices + allocator state as "
+                         "JSON and exit (debugging)")
     ap.add_argument("--exit-on-stream-loss", action="store_true",
                     help="exit(1) when a ListAndWatch stream breaks so the "
                          "DaemonSet restarts the pod (the ROCm plugin's "
@@ -96,6 +99,38 @@ def device_plugin_main(argv=None) -> int:
     except DriverUnavailableError as e:
         log.error("%s (exit 2)", e)
         return 2
+
+    if args.dump:
+        import json
+
+        from .allocator import AllocationError, BestEffortPolicy
+        from .topology import KFDTopology
+
+        topo = KFDTopology.load(paths)
+        out = {
+            "devices": {i: d.as_dict() for i, d in devices.items()},
+            "homogeneous": None,
+            "allocator": None,
+        }
+        from .topology.discovery import is_homogeneous, unique_partition_config_count
+
+        out["homogeneous"] = is_homogeneous(devices)
+        out["partition_configs"] = unique_partition_config_count(devices)
+        policy = BestEffortPolicy()
+        try:
+            policy.init([d for d in devices.values() if d.kfd_backed],
+                        topology=topo)
+            groups, node_of_id, weights = policy.export_state()
+            out["allocator"] = {
+                "groups": [{"parent": g[0], "node_ids": g[1]} for g in groups],
+                "pair_weights": [
+                    {"from": a, "to": b, "weight": w} for a, b, w in weights
+                ],
+            }
+        except AllocationError as e:
+            out["allocator"] = {"error": str(e)}
+        print(json.dumps(out, indent=2))
+        return 0
 
     try:
         resources = get_resource_list(devices, strategy)

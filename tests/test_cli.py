@@ -81,3 +81,18 @@ def test_cli_no_driver_exit2(tmp_path):
         cli_mod.time.monotonic = orig_mono
         cli_mod.time.sleep = orig_sleep
     assert rc == 2
+
+
+def test_cli_dump(tmp_path, fake_mi355x_8, capsys):
+    import json
+
+    from k8s_device_plugin_amd.cli import device_plugin_main
+
+    rc = device_plugin_main(["--dump", "--sysroot", fake_mi355x_8.paths.root])
+    assert rc == 0
+    out = json.loads(capsys.readouterr().out)
+    assert len(out["devices"]) == 8
+    assert out["homogeneous"] is True
+    assert out["partition_configs"] == {"spx_nps1": 8}
+    assert len(out["allocator"]["groups"]) == 8
+    assert len(out["allocator"]["pair_weights"]) == 28
