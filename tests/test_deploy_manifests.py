@@ -1,0 +1,59 @@
+"""Deployment manifests stay consistent with the actual CLI surface."""
+
+import glob
+import os
+
+import yaml
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _cli_flags(main):
+    import argparse
+
+    captured = {}
+    orig = argparse.ArgumentParser.parse_args
+
+    def fake(self, argv=None):
+        captured["parser"] = self
+        raise SystemExit(0)
+
+    argparse.ArgumentParser.parse_args = fake
+    try:
+        main(["--help-collect"])
+    except SystemExit:
+        pass
+    finally:
+        argparse.ArgumentParser.parse_args = orig
+    flags = set()
+    for action in captured["parser"]._actions:
+        flags.update(action.option_strings)
+    return flags
+
+
+def test_daemonset_args_exist_in_cli():
+    from k8s_device_plugin_amd.cli import device_plugin_main, labeller_main
+
+    dp_flags = _cli_flags(device_plugin_main)
+    nl_flags = _cli_flags(labeller_main)
+
+    for path in glob.glob(os.path.join(REPO, "deploy", "k8s-ds-*.yaml")):
+        for doc in yaml.safe_load_all(open(path)):
+            if not doc or doc.get("kind") != "DaemonSet":
+                continue
+            for c in doc["spec"]["template"]["spec"].get("containers", []):
+                flags = nl_flags if "labeller" in path else dp_flags
+                for arg in c.get("args", []):
+                    if arg.startswith("-"):
+                        assert arg in flags, f"{path}: unknown flag {arg}"
+
+
+def test_helm_values_match_labeller_flags():
+    from k8s_device_plugin_amd.labeller.labels import LABEL_KINDS
+
+    values = yaml.safe_load(
+        open(os.path.join(REPO, "deploy", "helm", "amd-gpu", "values.yaml"))
+    )
+    valid = {k.replace("-", "_") for k in LABEL_KINDS}
+    for flag in values["labeller"]["labels"]:
+        assert flag in valid, f"values.yaml labeller flag {flag} unknown"
