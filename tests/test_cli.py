@@ -96,3 +96,35 @@ def test_cli_dump(tmp_path, fake_mi355x_8, capsys):
     assert out["partition_configs"] == {"spx_nps1": 8}
     assert len(out["allocator"]["groups"]) == 8
     assert len(out["allocator"]["pair_weights"]) == 28
+
+
+def test_cli_mixed_strategy_cpx(tmp_path, fake_mi355x_cpx):
+    """Mixed naming on a CPX node: the daemon advertises amd.com/cpx_nps2
+    with all 64 partitions."""
+    from k8s_device_plugin_amd.testing.stub_kubelet import StubKubelet
+    from k8s_device_plugin_amd.protos import deviceplugin as dp
+
+    dp_dir = str(tmp_path / "dp")
+    kubelet = StubKubelet(dp_dir).start()
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "k8s_device_plugin_amd.cli",
+         "--resource_naming_strategy", "mixed",
+         "--kubelet-dir", dp_dir,
+         "--sysroot", fake_mi355x_cpx.paths.root],
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+    )
+    try:
+        reg = kubelet.wait_for_registration(timeout=30)
+        assert reg.resource_name == "amd.com/cpx_nps2"
+        stub = kubelet.connect(reg.endpoint)
+        call = stub.ListAndWatch(dp.Empty())
+        first = next(iter(call))
+        assert len(first.devices) == 64
+        call.cancel()
+        proc.send_signal(signal.SIGTERM)
+        assert proc.wait(timeout=15) == 0
+    finally:
+        if proc.poll() is None:
+            proc.kill()
+            proc.wait()
+        kubelet.stop()
