@@ -15,7 +15,6 @@ from __future__ import annotations
 
 import logging
 import os
-import re
 from typing import Callable, Dict, Optional
 
 from ..native import load_drmctl
@@ -115,9 +114,6 @@ def _physical_gpus(devices: Dict[str, GPUDevice]):
 
 def _sysfs_card_read(paths: SysPaths, card: int, *rel: str) -> Optional[str]:
     return read_stripped(os.path.join(paths.drm_card_device(card), *rel))
-
-
-_name_replacer = re.compile(r"[ ()]")
 
 
 def _sanitize_name(name: str) -> str:
