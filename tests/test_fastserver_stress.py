@@ -136,3 +136,42 @@ def test_restart_server_same_socket(tmp_path, fake_mi355x_8):
         assert stub.GetDevicePluginOptions(dp.Empty(), timeout=5) is not None
         ch.close()
         srv.stop()
+
+
+def test_large_request_multi_frame(native):
+    """A >100 KB Allocate request spans many DATA frames and needs
+    connection/stream window updates — the server must reassemble it."""
+    plugin, _, sock = native
+    ch = grpc.insecure_channel(f"unix://{sock}")
+    stub = dp.DevicePluginStub(ch)
+    ids = sorted(plugin.devices)
+    req = dp.AllocateRequest()
+    # 1500 containers x ~80 B each ~= 120 KB on the wire
+    for i in range(1500):
+        cr = req.container_requests.add()
+        cr.devices_ids.append(ids[i % len(ids)])
+        cr.devices_ids.append(f"pad-{'x' * 64}-{i}")  # unknown: skipped
+    assert len(req.SerializeToString()) > 100_000
+    resp = stub.Allocate(req, timeout=30)
+    assert len(resp.container_responses) == 1500
+    for car in resp.container_responses:
+        # kfd + 2 nodes for the one known device; unknown id skipped
+        assert len(car.devices) == 3
+    ch.close()
+
+
+def test_large_response_multi_frame(native):
+    """A large response (1500 containers x 3 specs) must stream out through
+    the data provider across many frames."""
+    plugin, _, sock = native
+    ch = grpc.insecure_channel(f"unix://{sock}")
+    stub = dp.DevicePluginStub(ch)
+    ids = sorted(plugin.devices)
+    req = dp.AllocateRequest()
+    for i in range(1500):
+        req.container_requests.add().devices_ids.extend(ids)
+    resp = stub.Allocate(req, timeout=30)
+    assert len(resp.container_responses) == 1500
+    assert len(resp.container_responses[-1].devices) == 1 + 2 * len(ids)
+    assert len(resp.SerializeToString()) > 500_000
+    ch.close()
