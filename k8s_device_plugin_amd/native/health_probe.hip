@@ -56,23 +56,28 @@ __device__ inline __bf16 small_int_bf16(int v) {
     return (__bf16)(float)v;  // small ints are exact in bf16
 }
 
-// One wave computes D = A*B with v_mfma_f32_16x16x32_bf16 twice:
+// One wave computes D = A*B with v_mfma_f32_16x16x32_bf16 three times:
 //  pass 0: A == 1, B == 0.5     -> every D element must equal 16.0
 //  pass 1: A[lane][r] = pattern, B == 1 -> sum(D) == 16 * sum(A) (exact)
-__global__ void mfma_probe_kernel(float *d_out /* [2][256] */) {
+//  pass 2: asymmetric A AND B   -> element-wise check against a host
+//          matmul through the documented lane maps (tests/test_gpu.py)
+__global__ void mfma_probe_kernel(float *d_out /* [3][256] */) {
 #if defined(__gfx950__)
     int lane = threadIdx.x;
     if (lane >= 64) return;
 
-    for (int pass = 0; pass < 2; ++pass) {
+    for (int pass = 0; pass < 3; ++pass) {
         bf16x8 a, b;
         for (int r = 0; r < 8; ++r) {
             if (pass == 0) {
                 a[r] = small_int_bf16(1);
                 b[r] = (__bf16)0.5f;
-            } else {
+            } else if (pass == 1) {
                 a[r] = small_int_bf16(((lane * 8 + r) % 7) - 3);
                 b[r] = small_int_bf16(1);
+            } else {
+                a[r] = small_int_bf16(((lane * 8 + r) % 7) - 3);
+                b[r] = small_int_bf16(((lane * 5 + r * 3) % 11) - 5);
             }
         }
         f32x4 c = {0.f, 0.f, 0.f, 0.f};
@@ -204,11 +209,11 @@ py::dict run_probe(int device, size_t hbm_bytes) {
     // 2. MFMA
     {
         float *d;
-        HIP_CHECK(hipMalloc(&d, 2 * 256 * sizeof(float)));
-        HIP_CHECK(hipMemset(d, 0, 2 * 256 * sizeof(float)));
+        HIP_CHECK(hipMalloc(&d, 3 * 256 * sizeof(float)));
+        HIP_CHECK(hipMemset(d, 0, 3 * 256 * sizeof(float)));
         hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, 0, d);
         HIP_CHECK(hipGetLastError());
-        std::vector<float> h(2 * 256);
+        std::vector<float> h(3 * 256);
         HIP_CHECK(hipMemcpy(h.data(), d, h.size() * sizeof(float),
                             hipMemcpyDeviceToHost));
         HIP_CHECK(hipFree(d));
@@ -327,6 +332,31 @@ py::dict run_probe(int device, size_t hbm_bytes) {
     return result;
 }
 
+// Raw MFMA fragment outputs for element-wise verification against a host
+// (PyTorch) reference: pass 0 = constant inputs, pass 1 = patterned A with
+// B == 1 (tests/test_gpu.py reconstructs the lane->element mapping).
+py::dict mfma_probe_raw(int device) {
+    HIP_CHECK(hipSetDevice(device));
+    float *d;
+    HIP_CHECK(hipMalloc(&d, 3 * 256 * sizeof(float)));
+    HIP_CHECK(hipMemset(d, 0, 3 * 256 * sizeof(float)));
+    hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, 0, d);
+    HIP_CHECK(hipGetLastError());
+    std::vector<float> h(3 * 256);
+    HIP_CHECK(hipMemcpy(h.data(), d, h.size() * sizeof(float),
+                        hipMemcpyDeviceToHost));
+    HIP_CHECK(hipFree(d));
+    py::list d0, d1, d2;
+    for (int i = 0; i < 256; ++i) d0.append(h[i]);
+    for (int i = 0; i < 256; ++i) d1.append(h[256 + i]);
+    for (int i = 0; i < 256; ++i) d2.append(h[512 + i]);
+    py::dict out;
+    out["pass0"] = d0;  // expect all 16.0
+    out["pass1"] = d1;  // D = A_pattern @ ones, per-lane fragments
+    out["pass2"] = d2;  // D = A_pattern @ B_pattern (asymmetric both)
+    return out;
+}
+
 int device_count() {
     int n = 0;
     hipError_t e = hipGetDeviceCount(&n);
@@ -392,4 +422,5 @@ PYBIND11_MODULE(_healthprobe, m) {
     m.def("device_count", &device_count);
     m.def("p2p_bandwidth", &p2p_bandwidth, py::arg("src") = 0,
           py::arg("dst") = 1, py::arg("bytes") = (size_t)1 << 30);
+    m.def("mfma_probe_raw", &mfma_probe_raw, py::arg("device") = 0);
 }
