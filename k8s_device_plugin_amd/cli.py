@@ -200,6 +200,10 @@ def labeller_main(argv=None) -> int:
                     help="k8s API base URL override (tests)")
     ap.add_argument("--oneshot", action="store_true",
                     help="reconcile once and exit (no watch)")
+    ap.add_argument("--refresh-interval", type=int, default=0,
+                    help="recompute labels every N seconds and reconcile on "
+                         "change (0 = compute once at startup, like the "
+                         "ROCm labeller)")
     ap.add_argument("-v", "--verbose", action="count", default=0)
     args = ap.parse_args(argv)
     _setup_logging(args.verbose)
@@ -228,6 +232,22 @@ def labeller_main(argv=None) -> int:
     if args.oneshot:
         ctl.reconcile()
         return 0
+
+    if args.refresh_interval > 0:
+        def refresher():
+            while not stop.wait(args.refresh_interval):
+                try:
+                    fresh = generate_labels(enabled, paths)
+                    patch = ctl.update_labels(fresh)
+                    if patch:
+                        log.info("labels changed; applied %d update(s)",
+                                 len(patch))
+                except Exception:
+                    log.exception("label refresh failed")
+
+        threading.Thread(target=refresher, daemon=True,
+                         name="label-refresh").start()
+
     ctl.run(block=True)
     return 0
 

@@ -29,6 +29,15 @@ class NodeLabelController:
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
 
+    def update_labels(self, labels: Dict[str, str]) -> Dict[str, Optional[str]]:
+        """Swap in freshly computed labels and reconcile if they changed
+        (used by the optional --refresh-interval loop; the reference
+        computes labels once per process lifetime, main.go:432)."""
+        if labels == self.labels:
+            return {}
+        self.labels = labels
+        return self.reconcile()
+
     def reconcile(self) -> Dict[str, Optional[str]]:
         """One reconcile pass; returns the patch that was applied."""
         node = self.client.get_node(self.node_name)

@@ -218,3 +218,33 @@ def test_watch_reconnects_after_stream_end(fake_mi355x_8):
         ctl.stop()
     finally:
         fake.stop()
+
+
+def test_update_labels_reconciles_on_change(fake_mi355x_8):
+    """update_labels (the --refresh-interval path) patches only when the
+    computed labels actually changed."""
+    import os
+
+    fake = FakeK8s(node_name="n3").start()
+    try:
+        labels = generate_labels({"vram": True}, fake_mi355x_8.paths)
+        client = K8sClient(base_url=fake.base_url)
+        ctl = NodeLabelController(client, "n3", labels)
+        ctl.reconcile()
+        assert fake.labels["amd.com/gpu.vram"] == "288G"
+
+        # unchanged -> no patch
+        assert ctl.update_labels(dict(labels)) == {}
+
+        # hardware changed (vram shrank): refresh must reconcile
+        node_dir = os.path.join(fake_mi355x_8.paths.kfd_topology_nodes, "2")
+        bank = os.path.join(node_dir, "mem_banks", "0", "properties")
+        text = open(bank).read().replace("309220868096", "154610434048")
+        open(bank, "w").write(text)
+        fresh = generate_labels({"vram": True}, fake_mi355x_8.paths)
+        patch = ctl.update_labels(fresh)
+        assert patch
+        assert fake.labels["beta.amd.com/gpu.vram.288G"] == "7"
+        assert fake.labels["beta.amd.com/gpu.vram.144G"] == "1"
+    finally:
+        fake.stop()
