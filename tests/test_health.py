@@ -65,3 +65,46 @@ def test_slow_exporter_falls_back_within_timeout(tmp_path):
         assert devs[0].health == "Healthy"  # fell back to node default
     finally:
         ex.stop()
+
+
+def test_exporter_get_gpu_state_stub(tmp_path):
+    """The GetGPUState RPC (defined for wire parity; the plugin itself uses
+    List, like the reference) round-trips against the fake exporter."""
+    import grpc
+
+    from k8s_device_plugin_amd.protos import metricssvc as ms
+
+    sock = str(tmp_path / "g.sock")
+    ex = FakeExporter(sock).start()
+    try:
+        ex.set_health("0000:0c:00.0", "unhealthy")
+        with grpc.insecure_channel(f"unix://{sock}") as ch:
+            stub = ms.MetricsServiceStub(ch)
+            resp = stub.GetGPUState(ms.GPUGetRequest(ID=["0"]), timeout=5)
+            assert resp.GPUState[0].Device == "0000:0c:00.0"
+            assert resp.GPUState[0].Health == "unhealthy"
+    finally:
+        ex.stop()
+
+
+def test_heartbeat_ticker_survives_bad_callback():
+    from k8s_device_plugin_amd.health import HeartbeatTicker
+    import time
+
+    hits = []
+    t = HeartbeatTicker(0.05)
+    t.subscribe(lambda: (_ for _ in ()).throw(RuntimeError("boom")))
+    t.subscribe(lambda: hits.append(1))
+    t.start()
+    time.sleep(0.4)
+    t.stop()
+    assert len(hits) >= 3, "good callback starved by a failing one"
+
+
+def test_heartbeat_ticker_zero_pulse_noop():
+    from k8s_device_plugin_amd.health import HeartbeatTicker
+
+    t = HeartbeatTicker(0)
+    t.start()
+    assert t._thread is None
+    t.stop()
