@@ -71,6 +71,17 @@ def main() -> int:
         if cuda:
             torch.cuda.synchronize()
 
+    probe_summary = None
+    if args.hbm_probe and cuda and rank == 0:
+        from k8s_device_plugin_amd.native import deep_health_probe
+
+        p = deep_health_probe(device=local_rank, hbm_bytes=1 << 30)
+        probe_summary = {
+            "healthy": p["healthy"],
+            "hbm_gbps": round(p["hbm_gbps"], 1),
+            "mfma_tflops": round(p["mfma_tflops"], 1),
+        }
+
     result = {}
     harness = None
     if rank == 0:
@@ -143,6 +154,7 @@ def main() -> int:
                 "preferred_alloc_p50_us": (
                     round(pref_p50_us, 1) if pref_p50_us is not None else None
                 ),
+                "deep_probe": probe_summary,
             },
         }
         print(json.dumps(out), flush=True)

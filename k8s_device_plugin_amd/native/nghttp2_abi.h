@@ -127,7 +127,6 @@ struct NgHttp2 {
     void (*session_del)(nghttp2_session *);
     ssize_t (*session_mem_recv)(nghttp2_session *, const uint8_t *, size_t);
     ssize_t (*session_mem_send)(nghttp2_session *, const uint8_t **);
-    int (*session_want_read)(nghttp2_session *);
     int (*session_want_write)(nghttp2_session *);
     int (*submit_settings)(nghttp2_session *, uint8_t,
                            const nghttp2_settings_entry *, size_t);
@@ -136,7 +135,6 @@ struct NgHttp2 {
     int (*submit_trailer)(nghttp2_session *, int32_t, const nghttp2_nv *,
                           size_t);
     int (*session_resume_data)(nghttp2_session *, int32_t);
-    int (*submit_rst_stream)(nghttp2_session *, uint8_t, int32_t, uint32_t);
     // client side (used by the native bench client)
     int (*session_client_new)(nghttp2_session **,
                               const nghttp2_session_callbacks *, void *);
@@ -178,13 +176,11 @@ struct NgHttp2 {
         LOAD(session_del, "nghttp2_session_del");
         LOAD(session_mem_recv, "nghttp2_session_mem_recv");
         LOAD(session_mem_send, "nghttp2_session_mem_send");
-        LOAD(session_want_read, "nghttp2_session_want_read");
         LOAD(session_want_write, "nghttp2_session_want_write");
         LOAD(submit_settings, "nghttp2_submit_settings");
         LOAD(submit_response, "nghttp2_submit_response");
         LOAD(submit_trailer, "nghttp2_submit_trailer");
         LOAD(session_resume_data, "nghttp2_session_resume_data");
-        LOAD(submit_rst_stream, "nghttp2_submit_rst_stream");
         LOAD(session_client_new, "nghttp2_session_client_new");
         LOAD(submit_request, "nghttp2_submit_request");
 #undef LOAD

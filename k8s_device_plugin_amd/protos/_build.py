@@ -27,11 +27,8 @@ _SCALAR_TYPES = {
     "bytes": F.TYPE_BYTES,
 }
 
-# Field spec: (name, number, type, repeated)
-# type: scalar name above, "msg:<fully.qualified.Type>", or "map<string,string>"
-FieldSpec = Tuple[str, int, str]
-
-
+# Field spec: (name, number, type[, "repeated"]); type is a scalar name
+# above, "msg:<fully.qualified.Type>", or "map<string,string>"
 class FileBuilder:
     def __init__(self, name: str, package: str, dependencies: Sequence[str] = ()):
         self.fdp = descriptor_pb2.FileDescriptorProto()
