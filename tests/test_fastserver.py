@@ -279,3 +279,31 @@ def test_stream_and_unary_same_connection(native):
     resp = stub.Allocate(req, timeout=5)
     assert len(resp.container_responses[0].devices) == 3
     call.cancel()
+
+
+def test_unknown_fields_tolerated(native):
+    """A newer kubelet may add fields to AllocateRequest; the native parser
+    must skip unknown varint/fixed/length-delimited fields."""
+    plugin, srv, _ = native
+    import grpc
+
+    ids = sorted(plugin.devices)
+    base = dp.AllocateRequest()
+    car = base.container_requests.add()
+    car.devices_ids.append(ids[0])
+    # append unknown fields at both levels:
+    #   container level: field 9 varint, field 10 bytes, field 11 fixed32
+    inner = car.SerializeToString() + b"\x48\x2a" + b"\x52\x03abc" + b"\x5d\x01\x02\x03\x04"
+    #   top level: rebuild with modified container + unknown field 7 bytes
+    raw = b"\x0a" + bytes([len(inner)]) + inner + b"\x3a\x04zzzz"
+
+    ch = grpc.insecure_channel(f"unix://{srv.socket_path}")
+    call = ch.unary_unary(
+        "/v1beta1.DevicePlugin/Allocate",
+        request_serializer=lambda b: b,
+        response_deserializer=dp.AllocateResponse.FromString,
+    )
+    resp = call(raw, timeout=5)
+    assert len(resp.container_responses) == 1
+    assert len(resp.container_responses[0].devices) == 3
+    ch.close()
