@@ -248,3 +248,41 @@ def test_mixed_links_prefer_xgmi(tmp_path):
     # 4-GPU request: the whole xGMI quad
     out = policy.allocate(ids[2:], [], 4)
     assert set(out) == set(ids[4:])
+
+
+def test_hive_and_partition_packing_combined(tmp_path):
+    """2 hives x 2 GPUs x 4 CPX partitions: an 8-partition request must
+    use the two GPUs of ONE hive, never straddle hives."""
+    fs = FakeSysfs(str(tmp_path / "hp"))
+    fs.add_cpu_node(0)
+    hives = [1111, 2222]
+    nodes = []
+    next_node, next_minor, next_xcp = 2, 136, 0
+    for g in range(4):
+        hive = hives[g // 2]
+        fs.add_physical_gpu(g, node_id=next_node, numa_node=0, hive_id=hive,
+                            compute_partition="CPX", memory_partition="NPS2")
+        nodes.append(next_node)
+        next_node += 1
+        for _ in range(3):
+            fs.add_partition(next_xcp, node_id=next_node, parent_index=g,
+                             card=8 + next_xcp, render_minor=next_minor,
+                             numa_node=0, hive_id=hive)
+            nodes.append(next_node)
+            next_node += 1
+            next_minor += 1
+            next_xcp += 1
+    for a in range(len(nodes)):
+        for b in range(a + 1, len(nodes)):
+            fs.add_link(nodes[a], nodes[b], link_type=11)
+
+    policy, devices = make_policy(fs)
+    ids = sorted(devices)
+    assert len(ids) == 16
+    out = policy.allocate(ids, [], 8)
+    assert len(out) == 8
+    # resolve each chosen device's hive via its topology node
+    topo = KFDTopology.load(fs.paths)
+    hives_used = {topo.nodes[devices[i].node_id].hive_id for i in out}
+    assert len(hives_used) == 1, f"request straddled hives: {out}"
+    assert len({devices[i].dev_id for i in out}) == 2
