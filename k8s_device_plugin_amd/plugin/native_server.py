@@ -84,9 +84,10 @@ class NativePluginServer:
         tables are rebuilt — the reference only refreshes its cache when
         the kubelet re-opens ListAndWatch (plugin.go:231)."""
         p = self.plugin
-        from ..topology import discover_gpus
+        from ..topology import KFDTopology, discover_gpus
 
-        fresh = discover_gpus(p.paths, strict=False)
+        topo = KFDTopology.load(p.paths)
+        fresh = discover_gpus(p.paths, topology=topo, strict=False)
         if set(fresh) != set(p.devices) or any(
             fresh[i].render_d != p.devices[i].render_d for i in fresh
         ):
@@ -120,7 +121,7 @@ class NativePluginServer:
                 except OSError as e:
                     log.warning("CDI spec refresh failed: %s", e)
 
-        devs = p.refreshed_device_list()
+        devs = p.refreshed_device_list(topology=topo)
         self._srv.push_list_update(
             dp.ListAndWatchResponse(devices=devs).SerializeToString()
         )

@@ -170,13 +170,16 @@ class AMDGPUPlugin:
             out.append(dev)
         return out
 
-    def refreshed_device_list(self) -> List:
+    def refreshed_device_list(self, topology=None) -> List:
         """Device list with freshly evaluated health: node-level kfd scan
         default, exporter per-GPU overrides, unbacked devices pinned
         Unhealthy.  Shared by the Python stream loop and the native
-        server's heartbeat push."""
+        server's heartbeat push (which passes its already-loaded topology
+        to avoid a second sysfs walk per beat)."""
         default = (
-            dp.HEALTHY if simple_health_check(self.paths) else dp.UNHEALTHY
+            dp.HEALTHY
+            if simple_health_check(self.paths, topology=topology)
+            else dp.UNHEALTHY
         )
         devs = self._device_list(health_default=default)
         populate_per_gpu_health(

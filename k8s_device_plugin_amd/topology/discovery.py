@@ -223,8 +223,10 @@ def count_gpus_from_topology(paths: SysPaths = SysPaths()) -> int:
     return sum(1 for n in topo.nodes.values() if n.simd_count > 0)
 
 
-def simple_health_check(paths: SysPaths = SysPaths()) -> bool:
+def simple_health_check(
+    paths: SysPaths = SysPaths(), topology: Optional[KFDTopology] = None
+) -> bool:
     """Node-level health: any kfd node with cpu_cores_count==0 and
     gfx_target_version>0 (reference: plugin.go:161-206)."""
-    topo = KFDTopology.load(paths)
+    topo = topology if topology is not None else KFDTopology.load(paths)
     return any(n.is_gpu for n in topo.nodes.values())
