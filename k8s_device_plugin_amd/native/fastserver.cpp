@@ -179,12 +179,14 @@ bool preferred_alloc(const AllocState &st,
             if (avail_nodes.count(n) && !req_nodes.count(n)) ids.push_back(n);
         if (!ids.empty()) groups.emplace_back(g.first, std::move(ids));
     }
-    std::sort(groups.begin(), groups.end(),
-              [](auto &a, auto &b) {
-                  if (a.second.size() != b.second.size())
-                      return a.second.size() < b.second.size();
-                  return a.first < b.first;
-              });
+    // stable: Python's sort is stable, and equal (size, parent) keys must
+    // tie-break identically for result parity
+    std::stable_sort(groups.begin(), groups.end(),
+                     [](auto &a, auto &b) {
+                         if (a.second.size() != b.second.size())
+                             return a.second.size() < b.second.size();
+                         return a.first < b.first;
+                     });
     if (groups.size() > 64) { err = "too many device groups"; return false; }
 
     int new_size = size - (int)req_node_list.size();
