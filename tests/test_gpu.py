@@ -176,19 +176,36 @@ def test_allocate_paths_are_real_devices(live_devices):
 
 
 def test_labeller_on_live_sysfs(live_devices):
-    """Label generation on the real MI355X: gfx950 values end-to-end."""
+    """Label generation on the real MI355X: gfx950 values end-to-end.
+
+    Partition-mode-dependent values (vram/cu/simd per schedulable device)
+    are derived from the live kfd tree so the test holds in SPX and CPX
+    alike; identity values are asserted fixed."""
     from k8s_device_plugin_amd.labeller import generate_labels
     from k8s_device_plugin_amd.labeller.labels import LABEL_KINDS
-    from k8s_device_plugin_amd.topology import SysPaths
+    from k8s_device_plugin_amd.topology import KFDTopology, SysPaths
 
+    paths, devs = live_devices
     labels = generate_labels({k: True for k in LABEL_KINDS}, SysPaths("/"))
-    assert labels["amd.com/gpu.vram"] == "288G"
-    assert labels["amd.com/gpu.cu-count"] == "256"
-    assert labels["amd.com/gpu.simd-count"] == "1024"
+
+    # identity facts: fixed for MI355X regardless of partition mode
     assert labels["amd.com/gpu.device-id"] == "75a3"
     assert labels["amd.com/gpu.family"] == "AI"
     assert "MI355" in labels["amd.com/gpu.product-name"]
-    assert labels["amd.com/gpu.compute-memory-partition"] == "spx_nps1"
+    assert labels["amd.com/gpu.compute-partitioning-supported"] in ("true", "false")
+
+    # mode-dependent values must match the first kfd-backed physical GPU
+    topo = KFDTopology.load(paths)
+    backed = next(d for d in devs.values() if d.kfd_backed and not d.is_partition)
+    node = topo.node_by_render_minor(backed.render_d)
+    g = round(node.vram_bytes // (1024 * 1024) / 1024)
+    assert labels["amd.com/gpu.vram"] == f"{g}G"
+    assert labels["amd.com/gpu.cu-count"] == str(node.cu_count)
+    assert labels["amd.com/gpu.simd-count"] == str(node.simd_count)
+    if backed.compute_partition == "spx":
+        assert labels["amd.com/gpu.vram"] == "288G"
+        assert labels["amd.com/gpu.cu-count"] == "256"
+        assert labels["amd.com/gpu.compute-memory-partition"].startswith("spx_")
     # firmware labels come from the raw-ioctl shim on a live box
     fw_labels = [k for k in labels if k.startswith("beta.amd.com/gpu.firmware.")]
     assert fw_labels, "expected firmware labels via drmctl ioctls"
