@@ -62,11 +62,12 @@ def test_advertised_vs_present(live_devices):
 
     _, devs = live_devices
     present = torch.cuda.device_count()
-    # schedulable (kfd-backed) physical GPUs must match what HIP sees;
+    # schedulable (kfd-backed) devices must match what HIP sees — whole
+    # GPUs in SPX, partitions in CPX (each partition is one HIP device);
     # cgroup-masked peers are advertised Unhealthy and don't count
-    physical = len({d.dev_id for d in devs.values() if d.kfd_backed})
-    assert physical == present, (
-        f"kfd walk found {physical} schedulable GPUs, torch sees {present}"
+    schedulable = sum(1 for d in devs.values() if d.kfd_backed)
+    assert schedulable == present, (
+        f"kfd walk found {schedulable} schedulable devices, torch sees {present}"
     )
 
 
