@@ -140,6 +140,13 @@ def main() -> int:
             "dtype": "n/a",
             "data": "synthetic",
             "config": {
+                # BASELINE.json's headline is a compound (Allocate p50 +
+                # advertised-vs-present); the scalar `value` is the
+                # admission aggregate and the compound's components follow
+                "baseline_metric": (
+                    "Allocate() p50 latency + advertised-vs-present GPUs "
+                    "at 1/2/4/8 MI355X"
+                ),
                 "model": "stub-kubelet-pod-admission",
                 "resource": "amd.com/gpu",
                 "devices_per_allocate": args.gpus,
