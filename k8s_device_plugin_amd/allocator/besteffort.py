@@ -63,6 +63,7 @@ class BestEffortPolicy:
         self._devices: Dict[str, GPUDevice] = {}
         self._weights: Dict[int, Dict[int, int]] = {}
         self._groups: Dict[str, _Group] = {}
+        self._initialized = False
 
     def init(
         self,
@@ -75,8 +76,21 @@ class BestEffortPolicy:
             raise AllocationError("no devices to initialize allocator with")
         topo = topology if topology is not None else KFDTopology.load(paths)
         self._weights = compute_pair_weights(devs, topo)
-        if not self._weights:
-            raise AllocationError("failed to initialize pair weights (no device links)")
+        if not self._weights and len(devs) > 1:
+            # Multiple devices but zero GPU-GPU links in the topology: the
+            # reference's Init fails here and the plugin silently drops
+            # GetPreferredAllocation (besteffort_policy.go:70-86,
+            # plugin.go:86-89).  We instead degrade to a zero-weight table:
+            # candidates are then ranked purely by the anti-fragmentation
+            # group ordering, which is still deterministic and correct —
+            # and the preferred-allocation path stays advertised (and
+            # measurable) instead of vanishing.
+            import logging
+
+            logging.getLogger(__name__).warning(
+                "no inter-device links in topology for %d devices; serving "
+                "preferred allocation with uniform weights", len(devs)
+            )
         self._devices = {d.id: d for d in devs}
         self._groups = {}
         for d in devs:
@@ -86,10 +100,11 @@ class BestEffortPolicy:
                 g.parent_id = d.id
         for g in self._groups.values():
             g.node_ids.sort()
+        self._initialized = True
 
     @property
     def initialized(self) -> bool:
-        return bool(self._weights)
+        return self._initialized
 
     def export_state(self):
         """State for the native fast server's in-C++ search: (groups as

@@ -60,6 +60,35 @@ _IDS_FILE = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file_
                          "native", "amdgpu.ids")
 
 
+def family_from_gfx_target(gfx: int) -> "str | None":
+    """sysfs fallback mapping: kfd gfx_target_version -> family name.
+
+    Used only when the DRM ioctl is unavailable; mirrors FAMILY_NAMES
+    coverage for every discrete-GPU generation rather than hardcoding one
+    branch, so a future-IP Instinct falls through to None (no label)
+    instead of silently mislabelling.  APU-specific families (KV/CZ/RV/
+    VGH/YC/GC_10_3_x) are indistinguishable from their discrete siblings
+    in kfd sysfs — those resolve correctly only via the ioctl path.
+    Encoding: major*10000 + minor*100 + rev (e.g. gfx950 -> 90500).
+    """
+    major, minor = gfx // 10000, (gfx // 100) % 100
+    if major == 6:
+        return "SI"
+    if major == 7:
+        return "CI"
+    if major == 8:
+        return "VI"
+    if major == 9:
+        return "AI"  # Vega + all CDNA incl. gfx942 (90402) and gfx950 (90500)
+    if major == 10:
+        return "NV"
+    if major == 11:
+        return "GC_11_5_0" if minor == 5 else "GC_11_0_0"
+    if major == 12:
+        return "GC_12_0_0"
+    return None
+
+
 def create_label_prefix(name: str, experimental: bool = False) -> str:
     prefix = EXPERIMENTAL_AMD_PREFIX if experimental else AMD_PREFIX
     return f"{prefix}/gpu.{name}"
@@ -167,10 +196,10 @@ def _gen_family(devices, paths, topo) -> Dict[str, str]:
             except RuntimeError:
                 pass
         if name is None and d.node_id in topo.nodes:
-            # sysfs fallback: gfx9xx targets are the AI (Vega-derived) family
+            # sysfs fallback keyed on gfx_target_version, covering every
+            # discrete generation in FAMILY_NAMES (gfx950 -> AI)
             gfx = topo.nodes[d.node_id].properties.get("gfx_target_version", 0)
-            if gfx >= 90000 and gfx < 100000:
-                name = "AI"
+            name = family_from_gfx_target(gfx)
         if name:
             counts[name] = counts.get(name, 0) + 1
     return _create_labels("family", counts) if counts else {}

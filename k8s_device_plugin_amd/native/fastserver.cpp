@@ -491,13 +491,41 @@ class Server {
     // ---------------- request dispatch ----------------
 
     void dispatch(Conn *conn, int32_t stream_id, Stream *st) {
+        // Validate the gRPC length-prefixed message framing before looking
+        // at the path.  A compressed-flag byte other than 0 means the
+        // client negotiated a message codec we do not implement
+        // (grpc-go only sets it after a grpc-encoding handshake) →
+        // UNIMPLEMENTED(12), matching grpc-go's own unsupported-codec
+        // status.  A body shorter than the declared length (truncated
+        // frame) or a stray partial prefix is a malformed request →
+        // INTERNAL(13).  Treating either as an empty request would turn
+        // garbage into a bogus success (e.g. an Allocate response with
+        // only /dev/kfd).
         std::string msg;
-        if (st->req_body.size() >= 5) {
+        if (!st->req_body.empty()) {
+            if (st->req_body.size() < 5) {
+                st->grpc_status = "13";  // INTERNAL: truncated frame prefix
+                st->grpc_message = "malformed grpc frame: short prefix";
+                submit_unary(conn, stream_id, st, "", false);
+                return;
+            }
+            uint8_t compressed = (uint8_t)st->req_body[0];
             uint32_t len;
             memcpy(&len, st->req_body.data() + 1, 4);
             len = ntohl(len);
-            if (st->req_body.size() >= 5 + len)
-                msg = st->req_body.substr(5, len);
+            if (compressed != 0) {
+                st->grpc_status = "12";  // UNIMPLEMENTED: compression
+                st->grpc_message = "grpc message compression not supported";
+                submit_unary(conn, stream_id, st, "", false);
+                return;
+            }
+            if (st->req_body.size() < (size_t)5 + len) {
+                st->grpc_status = "13";  // INTERNAL: truncated body
+                st->grpc_message = "malformed grpc frame: truncated body";
+                submit_unary(conn, stream_id, st, "", false);
+                return;
+            }
+            msg = st->req_body.substr(5, len);
         }
 
         const std::string &p = st->path;

@@ -169,14 +169,16 @@ class PluginManager:
             if cur is not None and cur != last:
                 log.info("kubelet.sock (re)created; re-registering all plugins")
                 for inst in list(self.plugins.values()):
-                    if inst.native:
-                        # the restarted kubelet will reopen ListAndWatch;
-                        # make sure it gets a fresh device/health snapshot
-                        # even when no pulse ticker runs
-                        try:
+                    # the restarted kubelet will reopen ListAndWatch; make
+                    # sure it gets a fresh device/health snapshot even when
+                    # no pulse ticker runs — both server implementations
+                    try:
+                        if inst.native:
                             inst.server.heartbeat()
-                        except Exception:
-                            log.exception("native state refresh failed")
+                        else:
+                            inst.plugin.heartbeat()
+                    except Exception:
+                        log.exception("state refresh failed")
                     self.register(inst)
             last = cur
 

@@ -77,6 +77,7 @@ class AMDGPUPlugin:
         self.exit_on_stream_loss = exit_on_stream_loss
         self._cond = threading.Condition()
         self._heartbeat_gen = 0
+        self._latest_list = None  # snapshot shared by all streams per beat
         self._stop = threading.Event()
 
     # ---- lifecycle ----
@@ -101,11 +102,56 @@ class AMDGPUPlugin:
 
     def stop(self) -> None:
         self._stop.set()
-        self.heartbeat()  # wake streams so they exit
+        with self._cond:
+            self._cond.notify_all()  # wake streams so they exit
+
+    def refresh_state(self):
+        """One shared refresh per heartbeat: a single sysfs walk + a single
+        exporter query, whose result every open ListAndWatch stream then
+        sends.  Before consolidation each stream independently re-walked
+        sysfs and re-dialed the exporter per beat (2 streams = 2 walks);
+        the native server already worked this way (native_server.py:79).
+
+        Also mirrors the native heartbeat's device-set tracking: if the
+        rediscovered set differs (hot-unplug, partition-mode change) the
+        allocator is re-inited so GetPreferredAllocation never serves
+        stale groups/weights (the reference has this gap — plugin.go:231
+        refreshes the cache but never re-runs allocator Init)."""
+        topo = KFDTopology.load(self.paths)
+        fresh = discover_gpus(self.paths, topology=topo, strict=False)
+        changed = set(fresh) != set(self.devices) or any(
+            fresh[i].render_d != self.devices[i].render_d for i in fresh
+        )
+        self.devices = fresh
+        if changed:
+            log.info(
+                "device set changed (%d devices); re-initializing allocator",
+                len(fresh),
+            )
+            schedulable = [d for d in fresh.values() if d.kfd_backed]
+            self.allocator_init_error = False
+            try:
+                self.allocator.init(schedulable, topology=topo)
+            except AllocationError as e:
+                log.error(
+                    "allocator re-init failed, falling back to kubelet "
+                    "default allocation: %s", e
+                )
+                self.allocator_init_error = True
+        return self.refreshed_device_list(topology=topo)
 
     def heartbeat(self) -> None:
-        """Trigger a health refresh on all ListAndWatch streams."""
+        """Refresh device/health state once and fan it out to all streams."""
+        if not self._stop.is_set():
+            try:
+                latest = self.refresh_state()
+            except Exception:
+                log.exception("heartbeat refresh failed; keeping last list")
+                latest = self._latest_list
+        else:
+            latest = self._latest_list
         with self._cond:
+            self._latest_list = latest
             self._heartbeat_gen += 1
             self._cond.notify_all()
 
@@ -227,7 +273,15 @@ class AMDGPUPlugin:
             if not fired:
                 continue
 
-            yield dp.ListAndWatchResponse(devices=self.refreshed_device_list())
+            # send the heartbeat's shared snapshot (one walk + one exporter
+            # query per beat, not per stream); fall back to computing our
+            # own only if a beat fired without a snapshot (direct
+            # heartbeat-gen bump in tests)
+            with self._cond:
+                latest = self._latest_list
+            if latest is None:
+                latest = self.refreshed_device_list()
+            yield dp.ListAndWatchResponse(devices=latest)
 
     def GetPreferredAllocation(self, request, context):
         import grpc
