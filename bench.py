@@ -39,6 +39,15 @@ def main() -> int:
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--hbm-probe", action="store_true",
                     help="also run the deep health probe before timing")
+    ap.add_argument("--curve", default=None, metavar="N1,N2,...",
+                    help="single-process scaling curve: measure each N "
+                         "(e.g. 1,2,4,8) back-to-back and report all of "
+                         "them in config.curve of the one JSON line")
+    ap.add_argument("--sysfs", choices=["auto", "live", "fake"],
+                    default="auto",
+                    help="topology source: auto = live kfd if present, "
+                         "else synthetic 8xMI355X tree; fake forces the "
+                         "synthetic tree (full 8-GPU curve on any box)")
     args = ap.parse_args()
 
     import torch
@@ -82,12 +91,22 @@ def main() -> int:
             "mfma_tflops": round(p["mfma_tflops"], 1),
         }
 
+    curve_ns = None
+    if args.curve:
+        curve_ns = sorted({int(x) for x in args.curve.split(",") if x.strip()})
+
     result = {}
     harness = None
     if rank == 0:
         try:
-            harness = _Harness(args.gpus)
+            harness = _Harness(
+                args.gpus if curve_ns is None else 1, sysfs=args.sysfs
+            )
             harness.start()
+            if curve_ns is not None:
+                # clamp to what the node actually advertises (a live box
+                # may expose fewer kfd-visible GPUs than requested)
+                curve_ns = [n for n in curve_ns if n <= harness.advertised]
             # warmup
             for _ in range(args.warmup):
                 harness.step()
@@ -99,6 +118,10 @@ def main() -> int:
             traceback.print_exc()
             print(f"bench harness failed: {e}", file=sys.stderr, flush=True)
             os._exit(1)
+
+    curve_out = None
+    if rank == 0 and curve_ns:
+        curve_out = _run_curve(harness, curve_ns, args.steps, args.warmup)
 
     barrier_sync()
     t0 = time.perf_counter()
@@ -162,6 +185,7 @@ def main() -> int:
                     round(pref_p50_us, 1) if pref_p50_us is not None else None
                 ),
                 "deep_probe": probe_summary,
+                "curve": curve_out,
             },
         }
         print(json.dumps(out), flush=True)
@@ -172,15 +196,49 @@ def main() -> int:
     return 0
 
 
+def _run_curve(harness: "_Harness", ns, steps: int, warmup: int):
+    """Measure each N back-to-back in one process: the 1/2/4/8 scaling
+    curve of the BASELINE headline (Allocate p50 + advertised-vs-present
+    at N GPUs requested), plus hive-packing evidence for the chosen sets."""
+    out = []
+    for n in ns:
+        harness.alloc_lat_us = []
+        harness.pref_lat_us = []
+        for _ in range(warmup):
+            harness.step(n)
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            harness.step(n)
+        t1 = time.perf_counter()
+        lat = sorted(harness.alloc_lat_us)
+        entry = {
+            "n": n,
+            "admissions_per_s": round(steps / (t1 - t0), 2),
+            "device_grants_per_s": round(n * steps / (t1 - t0), 2),
+            "ms_per_step": round((t1 - t0) / steps * 1e3, 4),
+            "allocate_p50_us": round(statistics.median(lat), 1),
+            "allocate_p99_us": round(lat[int(len(lat) * 0.99)], 1),
+            "preferred_alloc_p50_us": (
+                round(statistics.median(harness.pref_lat_us), 1)
+                if harness.pref_lat_us else None
+            ),
+            "hives_in_chosen_set": harness.hive_count(harness.last_chosen),
+        }
+        out.append(entry)
+    return out
+
+
 class _Harness:
     """Plugin + stub kubelet over live or synthetic sysfs (rank 0 only)."""
 
-    def __init__(self, n_gpus: int):
+    def __init__(self, n_gpus: int, sysfs: str = "auto"):
         self.n = n_gpus
+        self.sysfs_mode = sysfs
         self.alloc_lat_us = []
         self.pref_lat_us = []
         self.advertised = 0
         self.sysfs_kind = "unknown"
+        self.last_chosen = []
         self._tmp = None
         self._mgr = None
         self._kubelet = None
@@ -196,7 +254,14 @@ class _Harness:
         root = self._tmp.name
 
         live = SysPaths("/")
-        if os.path.isdir(live.kfd_class) and simple_health_check(live):
+        use_live = (
+            self.sysfs_mode != "fake"
+            and os.path.isdir(live.kfd_class)
+            and simple_health_check(live)
+        )
+        if self.sysfs_mode == "live" and not use_live:
+            raise RuntimeError("--sysfs live requested but no live kfd found")
+        if use_live:
             paths = live
             self.sysfs_kind = "live"
         else:
@@ -207,6 +272,7 @@ class _Harness:
             )
             paths = fs.paths
             self.sysfs_kind = "fake-8xMI355X"
+        self._paths = paths
 
         dp_dir = os.path.join(root, "device-plugins")
         self._kubelet = StubKubelet(dp_dir).start()
@@ -243,21 +309,24 @@ class _Harness:
                 f"advertised {self.advertised} healthy devices < requested {self.n}"
             )
 
-    def step(self) -> None:
+    def step(self, n: int = None) -> None:
         dp = self._dp
+        if n is None:
+            n = self.n
         # 1. GetPreferredAllocation for N devices (when advertised)
         if self.preferred_available:
             req = dp.PreferredAllocationRequest()
             cr = req.container_requests.add()
             cr.available_deviceIDs.extend(self.device_ids)
-            cr.allocation_size = self.n
+            cr.allocation_size = n
             t0 = time.perf_counter()
             resp = self._stub.GetPreferredAllocation(req, timeout=10)
             t1 = time.perf_counter()
             self.pref_lat_us.append((t1 - t0) * 1e6)
             chosen = list(resp.container_responses[0].deviceIDs)
         else:
-            chosen = self.device_ids[: self.n]
+            chosen = self.device_ids[:n]
+        self.last_chosen = chosen
 
         # 2. Allocate them
         areq = dp.AllocateRequest()
@@ -267,9 +336,28 @@ class _Harness:
         t1 = time.perf_counter()
         self.alloc_lat_us.append((t1 - t0) * 1e6)
         specs = aresp.container_responses[0].devices
-        assert len(specs) == 1 + 2 * self.n, (
+        assert len(specs) == 1 + 2 * n, (
             f"expected /dev/kfd + 2 nodes per device, got {len(specs)}"
         )
+
+    def hive_count(self, device_ids) -> int:
+        """Distinct xGMI hives spanned by a device-ID set (1 = packed)."""
+        try:
+            from k8s_device_plugin_amd.topology import (
+                KFDTopology, discover_gpus,
+            )
+
+            topo = KFDTopology.load(self._paths)
+            devs = discover_gpus(self._paths, topology=topo, strict=False)
+            hives = set()
+            for did in device_ids:
+                d = devs.get(did)
+                if d is None or d.node_id not in topo.nodes:
+                    return -1
+                hives.add(topo.nodes[d.node_id].hive_id)
+            return len(hives)
+        except Exception:
+            return -1
 
     def native_client_p50_us(self):
         """Allocate p50 measured with the C/nghttp2 bench client — the
