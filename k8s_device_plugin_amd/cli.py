@@ -252,8 +252,74 @@ def labeller_main(argv=None) -> int:
     return 0
 
 
+def partition_main(argv=None) -> int:
+    """Admin tool: show or set SPX/CPX + NPS partition modes.
+
+    Goes beyond the read-only reference (amdgpu.go:306-339); writes are
+    double-gated (--allow flag AND AMDXDP_ALLOW_REPARTITION=1 env) because
+    repartitioning tears down every kfd consumer on the node.
+    """
+    ap = argparse.ArgumentParser(
+        prog="amd-partitionctl",
+        description="Show or set AMD GPU compute/memory partition modes",
+    )
+    ap.add_argument("--sysroot", default="/", help="sysfs root override (tests)")
+    ap.add_argument("--compute", default=None,
+                    help="target compute mode (SPX/DPX/TPX/QPX/CPX)")
+    ap.add_argument("--memory", default=None,
+                    help="target memory mode (NPS1/NPS2/NPS4/NPS8)")
+    ap.add_argument("--gpus", default=None,
+                    help="comma-separated PCI addresses (default: all)")
+    ap.add_argument("--allow", action="store_true",
+                    help="confirm the destructive write (also requires "
+                         "AMDXDP_ALLOW_REPARTITION=1 in the environment; "
+                         "NEVER on shared boxes)")
+    ap.add_argument("--settle-timeout", type=float, default=30.0)
+    ap.add_argument("-v", "--verbose", action="count", default=0)
+    args = ap.parse_args(argv)
+    _setup_logging(args.verbose)
+
+    import json
+
+    from .topology import (
+        PartitionError,
+        available_partition_modes,
+        current_partition_modes,
+        set_partition_mode,
+    )
+
+    paths = SysPaths(args.sysroot)
+    if args.compute is None and args.memory is None:
+        avail_c, avail_m = available_partition_modes(paths)
+        print(json.dumps({
+            "current": {k: {"compute": c, "memory": m}
+                        for k, (c, m) in current_partition_modes(paths).items()},
+            "available_compute": avail_c,
+            "available_memory": avail_m,
+        }, indent=2))
+        return 0
+
+    try:
+        modes = set_partition_mode(
+            paths,
+            compute=args.compute,
+            memory=args.memory,
+            pci_addrs=args.gpus.split(",") if args.gpus else None,
+            allow=args.allow,
+            settle_timeout_s=args.settle_timeout,
+        )
+    except PartitionError as e:
+        print(f"error: {e}", file=sys.stderr)
+        return 1
+    print(json.dumps({k: {"compute": c, "memory": m}
+                      for k, (c, m) in modes.items()}, indent=2))
+    return 0
+
+
 if __name__ == "__main__":
     prog = os.path.basename(sys.argv[0])
     if "labeller" in prog or (len(sys.argv) > 1 and sys.argv[1] == "labeller"):
         sys.exit(labeller_main(sys.argv[2:] if sys.argv[1:2] == ["labeller"] else None))
+    if "partition" in prog or (len(sys.argv) > 1 and sys.argv[1] == "partition"):
+        sys.exit(partition_main(sys.argv[2:] if sys.argv[1:2] == ["partition"] else None))
     sys.exit(device_plugin_main())

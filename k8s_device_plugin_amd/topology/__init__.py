@@ -5,6 +5,13 @@ from .kfd import (
     parse_properties_file,
     parse_properties_text,
 )
+from .partition import (
+    PartitionError,
+    apply_partition_mode,
+    available_partition_modes,
+    current_partition_modes,
+    set_partition_mode,
+)
 from .discovery import (
     DriverUnavailableError,
     GPUDevice,
@@ -32,4 +39,9 @@ __all__ = [
     "is_memory_partition_supported",
     "count_gpus_from_topology",
     "simple_health_check",
+    "PartitionError",
+    "apply_partition_mode",
+    "available_partition_modes",
+    "current_partition_modes",
+    "set_partition_mode",
 ]

@@ -18,6 +18,7 @@ setup(
         "console_scripts": [
             "amd-device-plugin=k8s_device_plugin_amd.cli:device_plugin_main",
             "amd-node-labeller=k8s_device_plugin_amd.cli:labeller_main",
+            "amd-partitionctl=k8s_device_plugin_amd.cli:partition_main",
         ]
     },
 )
