@@ -55,6 +55,15 @@ def device_plugin_main(argv=None) -> int:
     ap.add_argument("--prestart-probe", action="store_true",
                     help="advertise pre_start_required and verify each "
                          "requested device answers before container start")
+    ap.add_argument("--prestart-deep", action="store_true",
+                    help="with --prestart-probe: additionally run the "
+                         "MFMA/LDS/HBM deep probe (with performance "
+                         "floors) on each requested GPU before start")
+    ap.add_argument("--deep-probe-every", type=int, default=0,
+                    help="run the deep GPU probe every Nth heartbeat and "
+                         "pin floor-failing GPUs Unhealthy (0 = off; "
+                         "floors via AMDXDP_MFMA_FLOOR_TFLOPS / "
+                         "AMDXDP_HBM_FLOOR_GBPS)")
     ap.add_argument("--dump", action="store_true",
                     help="print discovered devices + allocator state as "
                          "JSON and exit (debugging)")
@@ -155,6 +164,8 @@ def device_plugin_main(argv=None) -> int:
                                  cdi_enabled=args.cdi,
                                  cdi_spec_dir=args.cdi_dir,
                                  prestart_probe=args.prestart_probe,
+                                 prestart_deep=args.prestart_deep,
+                                 deep_probe_every=args.deep_probe_every,
                                  exit_on_stream_loss=args.exit_on_stream_loss),
         device_plugin_path=args.kubelet_dir or dp.DEVICE_PLUGIN_PATH,
         server_impl=args.server,

@@ -71,8 +71,62 @@ def load_fastserver(required: bool = False):
         return None
 
 
-def deep_health_probe(device: int = 0, hbm_bytes: int = 1 << 30) -> dict:
-    """Run the on-GPU MFMA/LDS/HBM probe.  Raises loudly when the extension
-    is missing on a GPU machine."""
+# Performance floors for the deep probe.  DVFS-aware defaults derived from
+# our own MI355X measurements (profiles/r01_bench_mi355x.md,
+# profiles/r01_pmc_mfma_peak.csv): healthy boxes measure 1935-2043 TF/s
+# bf16 MFMA (issue-saturated; DVFS clock-down explains the gap to the
+# 2495 TF/s zero-operand ceiling) and 6.2-6.3 TB/s HBM copy.  The floors
+# sit ~25% under the worst healthy observation so thermal jitter never
+# trips them but a silently degraded matrix pipe or downtrained HBM
+# channel does (before this, a degraded pipe still
+# reported healthy=true).  Overridable per deployment via env; 0 disables.
+DEFAULT_MFMA_FLOOR_TFLOPS = 1450.0
+DEFAULT_HBM_FLOOR_GBPS = 4700.0
+MFMA_FLOOR_ENV = "AMDXDP_MFMA_FLOOR_TFLOPS"
+HBM_FLOOR_ENV = "AMDXDP_HBM_FLOOR_GBPS"
+
+
+def _floor_from_env(env: str, default: float) -> float:
+    try:
+        return float(os.environ[env])
+    except (KeyError, ValueError):
+        return default
+
+
+def deep_health_probe(
+    device: int = 0,
+    hbm_bytes: int = 1 << 30,
+    mfma_floor_tflops: Optional[float] = None,
+    hbm_floor_gbps: Optional[float] = None,
+) -> dict:
+    """Run the on-GPU MFMA/LDS/HBM probe and apply performance floors.
+
+    Raises loudly when the extension is missing on a GPU machine.  The
+    result's `healthy` goes False on any correctness failure OR any floor
+    violation; violations are listed in `floor_violations`.
+    """
+    if mfma_floor_tflops is None:
+        mfma_floor_tflops = _floor_from_env(MFMA_FLOOR_ENV,
+                                            DEFAULT_MFMA_FLOOR_TFLOPS)
+    if hbm_floor_gbps is None:
+        hbm_floor_gbps = _floor_from_env(HBM_FLOOR_ENV,
+                                         DEFAULT_HBM_FLOOR_GBPS)
     mod = load_healthprobe(required=True)
-    return mod.run_probe(device=device, hbm_bytes=hbm_bytes)
+    res = mod.run_probe(device=device, hbm_bytes=hbm_bytes)
+    violations = []
+    mfma = res.get("mfma_tflops")
+    if mfma_floor_tflops > 0 and mfma is not None and mfma < mfma_floor_tflops:
+        violations.append(
+            f"mfma_tflops {mfma:.0f} < floor {mfma_floor_tflops:.0f}"
+        )
+    hbm = res.get("hbm_gbps")
+    if hbm_floor_gbps > 0 and hbm is not None and hbm < hbm_floor_gbps:
+        violations.append(f"hbm_gbps {hbm:.0f} < floor {hbm_floor_gbps:.0f}")
+    res["floors"] = {
+        "mfma_tflops": mfma_floor_tflops,
+        "hbm_gbps": hbm_floor_gbps,
+    }
+    res["floor_violations"] = violations
+    if violations:
+        res["healthy"] = False
+    return res

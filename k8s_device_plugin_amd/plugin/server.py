@@ -56,6 +56,8 @@ class AMDGPUPlugin:
         cdi_enabled: bool = False,
         cdi_spec_dir: Optional[str] = None,
         prestart_probe: bool = False,
+        prestart_deep: bool = False,
+        deep_probe_every: int = 0,
         dev_root: str = "/dev",
     ):
         self.resource = resource
@@ -66,6 +68,16 @@ class AMDGPUPlugin:
         # PreStartContainer is a no-op and never advertised,
         # plugin.go:219-224)
         self.prestart_probe = prestart_probe
+        # additionally run the full MFMA/LDS/HBM probe (with performance
+        # floors) before the container starts — catches a GPU whose device
+        # node answers but whose matrix pipe or HBM is silently degraded
+        self.prestart_deep = prestart_deep
+        # every Nth heartbeat, deep-probe each physical GPU and pin the
+        # whole GPU (all partitions) Unhealthy on correctness or floor
+        # failure; 0 disables (reference has no deep health at all)
+        self.deep_probe_every = deep_probe_every
+        self._deep_beat = 0
+        self._deep_failed: set = set()  # dev_ids that failed the deep probe
         self.dev_root = dev_root
         self.paths = paths
         self.devices: Dict[str, GPUDevice] = {}
@@ -138,7 +150,50 @@ class AMDGPUPlugin:
                     "default allocation: %s", e
                 )
                 self.allocator_init_error = True
+        if self.deep_probe_every > 0:
+            self._deep_beat += 1
+            if self._deep_beat % self.deep_probe_every == 0:
+                self._run_deep_check()
         return self.refreshed_device_list(topology=topo)
+
+    # ---- deep GPU health (beyond the reference) ----
+
+    def _hip_ordinal(self, dev: GPUDevice) -> int:
+        """Best-effort HIP device ordinal for a plugin device: HIP
+        enumerates GPUs in PCI bus order, and dev_id is the PCI-ish
+        address shared by a GPU's partitions, so the sorted physical
+        dev_id rank is the ordinal."""
+        phys = sorted({d.dev_id for d in self.devices.values() if d.kfd_backed})
+        return phys.index(dev.dev_id)
+
+    def _run_deep_check(self) -> None:
+        """Deep-probe every physical GPU; failures pin the whole GPU
+        (all its partitions) Unhealthy until a later probe passes."""
+        from ..native import NativeExtensionMissing, deep_health_probe
+
+        seen = set()
+        for dev in sorted(self.devices.values(), key=lambda d: d.id):
+            if not dev.kfd_backed or dev.dev_id in seen:
+                continue
+            seen.add(dev.dev_id)
+            try:
+                res = deep_health_probe(device=self._hip_ordinal(dev))
+            except NativeExtensionMissing as e:
+                # deployment problem, not a device problem: scream, don't flap
+                log.error("deep probe unavailable: %s", e)
+                return
+            except Exception as e:
+                log.error("deep probe failed on %s: %s", dev.dev_id, e)
+                self._deep_failed.add(dev.dev_id)
+                continue
+            if res.get("healthy"):
+                self._deep_failed.discard(dev.dev_id)
+            else:
+                log.error(
+                    "deep probe UNHEALTHY on %s: violations=%s",
+                    dev.dev_id, res.get("floor_violations"),
+                )
+                self._deep_failed.add(dev.dev_id)
 
     def heartbeat(self) -> None:
         """Refresh device/health state once and fan it out to all streams."""
@@ -195,6 +250,22 @@ class AMDGPUPlugin:
                         grpc.StatusCode.FAILED_PRECONDITION,
                         f"device {dev_id} failed the pre-start probe",
                     )
+                if self.prestart_deep:
+                    from ..native import deep_health_probe
+
+                    res = deep_health_probe(device=self._hip_ordinal(dev))
+                    if not res.get("healthy"):
+                        import grpc
+
+                        log.error(
+                            "PreStartContainer: device %s failed the deep "
+                            "probe: %s", dev_id, res.get("floor_violations"),
+                        )
+                        context.abort(
+                            grpc.StatusCode.FAILED_PRECONDITION,
+                            f"device {dev_id} failed the deep pre-start "
+                            f"probe: {res.get('floor_violations')}",
+                        )
         return dp.PreStartContainerResponse()
 
     def _my_devices(self) -> List[GPUDevice]:
@@ -232,10 +303,15 @@ class AMDGPUPlugin:
             devs, default, self.exporter_socket, self.exporter_timeout
         )
         # an exporter verdict cannot resurrect a device the kfd topology
-        # does not back
+        # does not back, nor one whose physical GPU failed the deep probe
         unbacked = {d.id for d in self._my_devices() if not d.kfd_backed}
+        deep_failed = {
+            d.id
+            for d in self._my_devices()
+            if d.dev_id and d.dev_id in self._deep_failed
+        }
         for dev in devs:
-            if dev.ID in unbacked:
+            if dev.ID in unbacked or dev.ID in deep_failed:
                 dev.health = dp.UNHEALTHY
         return devs
 
