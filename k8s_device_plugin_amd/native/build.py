@@ -85,8 +85,22 @@ def build_fastserver(force: bool = False) -> str:
     return out
 
 
+def build_h2tool(force: bool = False) -> str:
+    src = os.path.join(PKG_DIR, "h2tool.cpp")
+    out = os.path.join(PKG_DIR, "_h2tool.so")
+    hdr = os.path.join(PKG_DIR, "nghttp2_abi.h")
+    if force or _needs_build(src, out) or _needs_build(hdr, out):
+        _run(
+            ["g++", "-O2", "-std=c++17", "-shared", "-fPIC", src, "-o", out,
+             "-ldl"]
+            + _python_includes()
+        )
+    return out
+
+
 def build_all(force: bool = False) -> list:
-    return [build_drmctl(force), build_healthprobe(force), build_fastserver(force)]
+    return [build_drmctl(force), build_healthprobe(force),
+            build_fastserver(force), build_h2tool(force)]
 
 
 if __name__ == "__main__":
