@@ -348,7 +348,10 @@ class Server {
             alloc_.weights[((uint64_t)(uint32_t)a << 32) | (uint32_t)b] =
                 std::get<2>(t);
         }
-        alloc_.ready = !alloc_.weights.empty();
+        // weights may legitimately be empty: on a no-links topology
+        // (1-kfd-visible box) the policy degrades to uniform weights and
+        // every pair lookup scores 0 — the search is still well-defined
+        alloc_.ready = !alloc_.groups.empty();
     }
 
     void start() {

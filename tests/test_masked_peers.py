@@ -66,6 +66,39 @@ def test_masked_peers_unhealthy(tmp_path):
     plugin.stop()
 
 
+def test_masked_peers_native_server_preferred_allocation(tmp_path):
+    """Regression (found on a live 1-kfd-visible MI355X box, r02): the
+    NATIVE server must also serve the uniform-weight preferred path — its
+    ready flag used to require a non-empty weight table, so the python
+    side advertised GetPreferredAllocation but the C++ search answered
+    INVALID_ARGUMENT 'allocator not initialized'."""
+    import grpc
+
+    from k8s_device_plugin_amd.plugin.native_server import NativePluginServer
+
+    fs = build_mi355x_node(str(tmp_path / "r"), n_gpus=8)
+    _mask_kfd_nodes(fs, {0, 1, 2})
+
+    plugin = AMDGPUPlugin(resource="gpu", paths=fs.paths)
+    plugin.start()
+    assert not plugin.allocator_init_error
+    srv = NativePluginServer(plugin, str(tmp_path / "s.sock"))
+    srv.start()
+    try:
+        ch = grpc.insecure_channel(f"unix://{tmp_path}/s.sock")
+        stub = dp.DevicePluginStub(ch)
+        backed = [d.id for d in plugin.devices.values() if d.kfd_backed]
+        req = dp.PreferredAllocationRequest()
+        cr = req.container_requests.add()
+        cr.available_deviceIDs.extend(backed)
+        cr.allocation_size = 1
+        resp = stub.GetPreferredAllocation(req, timeout=10)
+        assert list(resp.container_responses[0].deviceIDs) == backed
+        ch.close()
+    finally:
+        srv.stop()
+
+
 def test_masked_peers_heartbeat_keeps_unbacked_unhealthy(tmp_path):
     fs = build_mi355x_node(str(tmp_path / "r"), n_gpus=4)
     _mask_kfd_nodes(fs, {0, 1, 2, 3})  # two GPUs visible
