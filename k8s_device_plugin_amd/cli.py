@@ -67,6 +67,13 @@ def device_plugin_main(argv=None) -> int:
     ap.add_argument("--dump", action="store_true",
                     help="print discovered devices + allocator state as "
                          "JSON and exit (debugging)")
+    ap.add_argument("--dump-podresources", action="store_true",
+                    help="query the kubelet PodResources API and print "
+                         "which pods hold amd.com/* devices, then exit")
+    ap.add_argument("--podresources-socket",
+                    default=None,
+                    help="kubelet pod-resources socket (default "
+                         "/var/lib/kubelet/pod-resources/kubelet.sock)")
     ap.add_argument("--exit-on-stream-loss", action="store_true",
                     help="exit(1) when a ListAndWatch stream breaks so the "
                          "DaemonSet restarts the pod (the ROCm plugin's "
@@ -93,6 +100,20 @@ def device_plugin_main(argv=None) -> int:
         return 1
 
     paths = SysPaths(args.sysroot)
+
+    if args.dump_podresources:
+        import json
+
+        from .plugin.podresources import gpu_allocation_summary
+        from .protos.podresources import PODRESOURCES_SOCKET
+
+        sock = args.podresources_socket or PODRESOURCES_SOCKET
+        try:
+            print(json.dumps(gpu_allocation_summary(sock), indent=2))
+        except Exception as e:
+            log.error("podresources query failed: %s", e)
+            return 1
+        return 0
 
     # gate on the ROCm driver being present (reference: main.go:139-152)
     deadline = time.monotonic() + 60
