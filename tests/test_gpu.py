@@ -102,9 +102,19 @@ def test_deep_health_probe():
     assert probe["mfma_ok"], probe
     assert probe["lds_ok"], probe
     assert probe["hbm_copy_ok"], probe
+    # healthy now also implies the DVFS-aware performance floors held
+    # (1450 TF/s MFMA, 4700 GB/s HBM) — a healthy MI355X must clear them
+    # with margin, and the floors fields must be reported
     assert probe["healthy"]
-    # MI355X HBM3E: ~6.3 TB/s achievable; even a sick part should beat 1 TB/s
-    assert probe["hbm_gbps"] > 1000, probe
+    assert probe["floor_violations"] == [], probe
+    assert probe["floors"]["mfma_tflops"] > 0
+    assert probe["mfma_tflops"] > probe["floors"]["mfma_tflops"], probe
+    assert probe["hbm_gbps"] > probe["floors"]["hbm_gbps"], probe
+    # an absurd floor must flip the verdict on the same hardware
+    degraded = deep_health_probe(device=0, hbm_bytes=1 << 28,
+                                 mfma_floor_tflops=10_000_000)
+    assert not degraded["healthy"]
+    assert degraded["floor_violations"], degraded
 
 
 def test_smoke_entry():
@@ -365,3 +375,61 @@ def test_mfma_elementwise_vs_torch():
 
     # pass 0/1 invariants still hold on the raw path
     assert all(v == 16.0 for v in raw["pass0"])
+
+
+def test_partitionctl_read_live(live_devices):
+    """Read-only partition state on live sysfs (writes are forbidden on
+    shared boxes — the gate refuses without AMDXDP_ALLOW_REPARTITION)."""
+    import os
+
+    from k8s_device_plugin_amd.topology import (
+        PartitionError,
+        SysPaths,
+        available_partition_modes,
+        current_partition_modes,
+        set_partition_mode,
+    )
+
+    paths = SysPaths("/")
+    cur = current_partition_modes(paths)
+    assert cur, "no amdgpu PCI devices visible"
+    for addr, (comp, mem) in cur.items():
+        # live MI355X reports real modes (SPX/CPX..., NPS1...)
+        assert comp in ("SPX", "DPX", "TPX", "QPX", "CPX", ""), (addr, comp)
+        assert mem.startswith("NPS") or mem == "", (addr, mem)
+    comp_avail, mem_avail = available_partition_modes(paths)
+    # MI355X supports compute partitioning; the capability files exist
+    assert "SPX" in comp_avail or comp_avail == []
+    # the safety gate must refuse writes on this shared box
+    os.environ.pop("AMDXDP_ALLOW_REPARTITION", None)
+    try:
+        set_partition_mode(paths, compute="SPX", allow=True)
+        raise AssertionError("gate did not refuse")
+    except PartitionError:
+        pass
+
+
+def test_deep_probe_heartbeat_live(live_devices):
+    """--deep-probe-every on real hardware: the heartbeat deep check runs
+    the MFMA/LDS/HBM probe against the live GPU and keeps it Healthy."""
+    from k8s_device_plugin_amd.plugin import AMDGPUPlugin
+    from k8s_device_plugin_amd.protos import deviceplugin as dp
+    from k8s_device_plugin_amd.topology import SysPaths
+
+    class _Ctx:
+        def is_active(self):
+            return True
+
+    plugin = AMDGPUPlugin(resource="gpu", paths=SysPaths("/"),
+                          deep_probe_every=1)
+    plugin.start()
+    stream = plugin.ListAndWatch(dp.Empty(), _Ctx())
+    next(stream)
+    plugin.heartbeat()  # runs the real probe on every kfd-backed GPU
+    resp = next(stream)
+    backed = {d.id for d in plugin.devices.values() if d.kfd_backed}
+    health = {d.ID: d.health for d in resp.devices}
+    for did in backed:
+        assert health[did] == "Healthy", (did, plugin._deep_failed)
+    assert not plugin._deep_failed
+    plugin.stop()
