@@ -121,6 +121,7 @@ class NativePluginServer:
                 except OSError as e:
                     log.warning("CDI spec refresh failed: %s", e)
 
+        p.maybe_deep_check()  # --deep-probe-every applies to this path too
         devs = p.refreshed_device_list(topology=topo)
         self._srv.push_list_update(
             dp.ListAndWatchResponse(devices=devs).SerializeToString()

@@ -150,13 +150,20 @@ class AMDGPUPlugin:
                     "default allocation: %s", e
                 )
                 self.allocator_init_error = True
-        if self.deep_probe_every > 0:
-            self._deep_beat += 1
-            if self._deep_beat % self.deep_probe_every == 0:
-                self._run_deep_check()
+        self.maybe_deep_check()
         return self.refreshed_device_list(topology=topo)
 
     # ---- deep GPU health (beyond the reference) ----
+
+    def maybe_deep_check(self) -> None:
+        """Deep-probe cadence shared by both serving paths: every Nth
+        heartbeat when --deep-probe-every is set (called from the python
+        refresh and from the native server's heartbeat push)."""
+        if self.deep_probe_every <= 0:
+            return
+        self._deep_beat += 1
+        if self._deep_beat % self.deep_probe_every == 0:
+            self._run_deep_check()
 
     def _hip_ordinal(self, dev: GPUDevice) -> int:
         """Best-effort HIP device ordinal for a plugin device: HIP
@@ -169,6 +176,11 @@ class AMDGPUPlugin:
     def _run_deep_check(self) -> None:
         """Deep-probe every physical GPU; failures pin the whole GPU
         (all its partitions) Unhealthy until a later probe passes."""
+        if not os.path.exists(os.path.join(self.dev_root, "kfd")):
+            # fake-sysfs / CPU environment: there is no GPU to probe;
+            # skip rather than condemning every (synthetic) device
+            log.debug("deep probe skipped: %s/kfd not present", self.dev_root)
+            return
         from ..native import NativeExtensionMissing, deep_health_probe
 
         seen = set()

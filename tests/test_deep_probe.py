@@ -88,11 +88,20 @@ class _Ctx:
         return True
 
 
+def _dev_root(tmp_path):
+    d = tmp_path / "dev"
+    d.mkdir(exist_ok=True)
+    (d / "kfd").touch()
+    return str(d)
+
+
+
 def test_heartbeat_deep_check_flips_unhealthy(tmp_path, stub_probe):
     fs = build_mi355x_node(str(tmp_path / "n"), n_gpus=2)
     mod = stub_probe(_StubProbeMod(mfma=900))  # below floor
 
-    plugin = AMDGPUPlugin(resource="gpu", paths=fs.paths, deep_probe_every=2)
+    plugin = AMDGPUPlugin(resource="gpu", paths=fs.paths, deep_probe_every=2,
+                          dev_root=_dev_root(tmp_path))
     plugin.start()
     stream = plugin.ListAndWatch(dp.Empty(), _Ctx())
     first = next(stream)
@@ -128,7 +137,8 @@ def test_deep_check_pins_all_partitions_of_failed_gpu(tmp_path, stub_probe):
             return r
 
     stub_probe(_FirstGpuBad())
-    plugin = AMDGPUPlugin(resource="gpu", paths=fs.paths, deep_probe_every=1)
+    plugin = AMDGPUPlugin(resource="gpu", paths=fs.paths, deep_probe_every=1,
+                          dev_root=_dev_root(tmp_path))
     plugin.start()
     stream = plugin.ListAndWatch(dp.Empty(), _Ctx())
     next(stream)
@@ -174,3 +184,51 @@ def test_prestart_deep_aborts_on_floor_failure(tmp_path, stub_probe):
         plugin.PreStartContainer(req, ctx)
     assert ctx.code == grpc.StatusCode.FAILED_PRECONDITION
     plugin.stop()
+
+
+def test_deep_check_skipped_without_dev_kfd(tmp_path, stub_probe):
+    """CPU/fake environments (no /dev/kfd) must not condemn synthetic
+    devices just because the probe cannot run."""
+    fs = build_mi355x_node(str(tmp_path / "n"), n_gpus=2)
+    mod = stub_probe(_StubProbeMod(mfma=900))  # would fail floors IF run
+    plugin = AMDGPUPlugin(resource="gpu", paths=fs.paths, deep_probe_every=1,
+                          dev_root=str(tmp_path / "emptydev"))
+    plugin.start()
+    stream = plugin.ListAndWatch(dp.Empty(), _Ctx())
+    next(stream)
+    plugin.heartbeat()
+    resp = next(stream)
+    assert all(d.health == "Healthy" for d in resp.devices)
+    assert mod.calls == []  # probe never invoked
+    plugin.stop()
+
+
+def test_native_heartbeat_triggers_deep_check(tmp_path, stub_probe):
+    """--deep-probe-every must apply to the NATIVE serving path too (the
+    default daemon path; gap found by the r02 soak rehearsal)."""
+    import grpc
+
+    from k8s_device_plugin_amd.plugin.native_server import NativePluginServer
+
+    fs = build_mi355x_node(str(tmp_path / "n"), n_gpus=2)
+    mod = stub_probe(_StubProbeMod(mfma=900))  # below floor
+    plugin = AMDGPUPlugin(resource="gpu", paths=fs.paths, deep_probe_every=1,
+                          dev_root=_dev_root(tmp_path))
+    plugin.start()
+    srv = NativePluginServer(plugin, str(tmp_path / "s.sock"))
+    srv.start()
+    try:
+        ch = grpc.insecure_channel(f"unix://{tmp_path}/s.sock")
+        stub = dp.DevicePluginStub(ch)
+        call = stub.ListAndWatch(dp.Empty())
+        it = iter(call)
+        first = next(it)
+        assert all(d.health == "Healthy" for d in first.devices)
+        srv.heartbeat()
+        resp = next(it)
+        assert all(d.health == "Unhealthy" for d in resp.devices)
+        assert sorted(set(mod.calls)) == [0, 1]
+        call.cancel()
+        ch.close()
+    finally:
+        srv.stop()
