@@ -44,6 +44,15 @@ class _ManagerCollector:
             g.add_metric([resource], len(inst.plugin.devices))
         yield g
 
+        df = GaugeMetricFamily(
+            "amdgpu_dp_deep_probe_failed_gpus",
+            "physical GPUs currently pinned Unhealthy by the deep probe",
+            labels=["resource"],
+        )
+        for resource, inst in self.manager.plugins.items():
+            df.add_metric([resource], len(inst.plugin._deep_failed))
+        yield df
+
         counters = {}
         for resource, inst in self.manager.plugins.items():
             if inst.native:
