@@ -18,9 +18,13 @@ from typing import Iterable
 
 from ..topology.discovery import GPUDevice
 
-CDI_VERSION = "0.6.0"
+# CDI 0.7: annotations carry per-device topology
+# hints (NUMA node, partition config, physical-GPU id) so operators and
+# runtime hooks can read placement facts straight from the spec.
+CDI_VERSION = "0.7.0"
 CDI_KIND = "amd.com/gpu"
 CDI_SPEC_DIR = "/var/run/cdi"
+ANNOTATION_PREFIX = "cdi.amd.com"
 
 
 def cdi_device_name(device_id: str, kind: str = CDI_KIND) -> str:
@@ -31,6 +35,9 @@ def build_cdi_spec(devices: Iterable[GPUDevice], kind: str = CDI_KIND) -> dict:
     spec = {
         "cdiVersion": CDI_VERSION,
         "kind": kind,
+        "annotations": {
+            f"{ANNOTATION_PREFIX}/producer": "k8s-device-plugin-mi355x",
+        },
         # /dev/kfd is shared by every GPU workload on the node
         "containerEdits": {
             "deviceNodes": [{"path": "/dev/kfd", "permissions": "rw"}]
@@ -38,9 +45,16 @@ def build_cdi_spec(devices: Iterable[GPUDevice], kind: str = CDI_KIND) -> dict:
         "devices": [],
     }
     for d in sorted(devices, key=lambda x: x.id):
+        annotations = {
+            f"{ANNOTATION_PREFIX}/numa-node": str(d.numa_node),
+            f"{ANNOTATION_PREFIX}/physical-gpu": d.dev_id or d.id,
+        }
+        if d.compute_partition and d.memory_partition:
+            annotations[f"{ANNOTATION_PREFIX}/partition"] = d.partition_key
         spec["devices"].append(
             {
                 "name": d.id,
+                "annotations": annotations,
                 "containerEdits": {
                     "deviceNodes": [
                         {"path": f"/dev/dri/card{d.card}", "permissions": "rw"},

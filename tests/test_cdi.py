@@ -22,7 +22,12 @@ def test_cdi_spec_contents(tmp_path, fake_mi355x_8):
     path = write_cdi_spec(devices.values(), spec_dir=str(tmp_path / "cdi"))
     spec = json.load(open(path))
     assert spec["kind"] == "amd.com/gpu"
-    assert spec["cdiVersion"] == "0.6.0"
+    assert spec["cdiVersion"] == "0.7.0"
+    # 0.7 annotations: per-device topology hints
+    for dev in spec["devices"]:
+        ann = dev["annotations"]
+        assert "cdi.amd.com/numa-node" in ann
+        assert "cdi.amd.com/physical-gpu" in ann
     assert len(spec["devices"]) == 8
     assert spec["containerEdits"]["deviceNodes"][0]["path"] == "/dev/kfd"
     d0 = next(d for d in spec["devices"] if d["name"] == "0000:0c:00.0")
