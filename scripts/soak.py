@@ -189,6 +189,8 @@ def main() -> int:
         stats["errors"].append("daemon did not exit on SIGTERM")
 
     deep_lines = [l.strip() for l in daemon_log if "deep probe" in l.lower()]
+    err_lines = [l.strip() for l in daemon_log
+                 if "ERROR" in l or "Traceback" in l]
     cdi_ok = bool(
         [f for f in (os.listdir(cdi_dir) if os.path.isdir(cdi_dir) else [])]
     )
@@ -204,6 +206,9 @@ def main() -> int:
         "daemon_exit": daemon.returncode,
         "cdi_spec_written": cdi_ok,
         "deep_probe_log_lines": len(deep_lines),
+        "deep_probe_lines_sample": deep_lines[:6],
+        "daemon_error_lines": len(err_lines),
+        "daemon_error_sample": err_lines[:6],
         "metrics": metrics,
     }
     print(json.dumps(out, indent=2))
