@@ -103,7 +103,7 @@ def test_deep_health_probe():
     assert probe["lds_ok"], probe
     assert probe["hbm_copy_ok"], probe
     # healthy now also implies the DVFS-aware performance floors held
-    # (1450 TF/s MFMA, 4700 GB/s HBM) — a healthy MI355X must clear them
+    # (1000 TF/s MFMA, 3800 GB/s HBM) — a healthy MI355X must clear them
     # with margin, and the floors fields must be reported
     assert probe["healthy"]
     assert probe["floor_violations"] == [], probe
