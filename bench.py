@@ -48,6 +48,9 @@ def main() -> int:
                     help="topology source: auto = live kfd if present, "
                          "else synthetic 8xMI355X tree; fake forces the "
                          "synthetic tree (full 8-GPU curve on any box)")
+    ap.add_argument("--fake-partitions", type=int, default=1,
+                    help="with --sysfs fake: CPX-style partitions per GPU "
+                         "(8 -> 64 logical devices, BASELINE config 4)")
     args = ap.parse_args()
 
     import torch
@@ -100,7 +103,8 @@ def main() -> int:
     if rank == 0:
         try:
             harness = _Harness(
-                args.gpus if curve_ns is None else 1, sysfs=args.sysfs
+                args.gpus if curve_ns is None else 1, sysfs=args.sysfs,
+                fake_partitions=args.fake_partitions,
             )
             harness.start()
             if curve_ns is not None:
@@ -231,9 +235,11 @@ def _run_curve(harness: "_Harness", ns, steps: int, warmup: int):
 class _Harness:
     """Plugin + stub kubelet over live or synthetic sysfs (rank 0 only)."""
 
-    def __init__(self, n_gpus: int, sysfs: str = "auto"):
+    def __init__(self, n_gpus: int, sysfs: str = "auto",
+                 fake_partitions: int = 1):
         self.n = n_gpus
         self.sysfs_mode = sysfs
+        self.fake_partitions = fake_partitions
         self.alloc_lat_us = []
         self.pref_lat_us = []
         self.advertised = 0
@@ -268,10 +274,15 @@ class _Harness:
             from k8s_device_plugin_amd.testing.fakesysfs import build_mi355x_node
 
             fs = build_mi355x_node(
-                os.path.join(root, "fakesys"), n_gpus=max(self.n, 8)
+                os.path.join(root, "fakesys"), n_gpus=max(self.n, 8),
+                partitions_per_gpu=self.fake_partitions,
+                compute_partition="CPX" if self.fake_partitions > 1 else "SPX",
             )
             paths = fs.paths
-            self.sysfs_kind = "fake-8xMI355X"
+            self.sysfs_kind = (
+                "fake-8xMI355X" if self.fake_partitions <= 1
+                else f"fake-8xMI355X-cpx{self.fake_partitions}"
+            )
         self._paths = paths
 
         dp_dir = os.path.join(root, "device-plugins")
