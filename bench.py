@@ -185,6 +185,9 @@ def main() -> int:
                 "allocate_p50_us": round(alloc_p50_us, 1),
                 "allocate_p99_us": round(alloc_p99_us, 1),
                 "native_client_allocate_p50_us": native_client_p50,
+                "native_client_preferred_p50_us": getattr(
+                    harness, "native_pref_p50_us", None
+                ),
                 "preferred_alloc_p50_us": (
                     round(pref_p50_us, 1) if pref_p50_us is not None else None
                 ),
@@ -393,14 +396,15 @@ class _Harness:
         sock = self._mgr.plugins["gpu"].socket_path
         try:
             out = subprocess.run(
-                [exe, sock, self.device_ids[0], "2000"],
+                [exe, sock, self.device_ids[0], "2000",
+                 ",".join(self.device_ids), str(self.n)],
                 capture_output=True, text=True, timeout=120,
             )
             if out.returncode != 0:
                 return None
-            return json.loads(out.stdout.strip().splitlines()[-1])[
-                "allocate_p50_us"
-            ]
+            rec = json.loads(out.stdout.strip().splitlines()[-1])
+            self.native_pref_p50_us = rec.get("preferred_p50_us")
+            return rec["allocate_p50_us"]
         except Exception:
             return None
 

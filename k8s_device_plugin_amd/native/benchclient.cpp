@@ -110,6 +110,31 @@ std::string allocate_request_bytes(const std::string &device_id) {
     return framed;
 }
 
+std::string preferred_request_bytes(const std::vector<std::string> &ids,
+                                    int size) {
+    // ContainerPreferredAllocationRequest{available_deviceIDs: ids,
+    //                                     allocation_size: size}
+    std::string cr;
+    for (auto &id : ids) {
+        cr.push_back('\x0a');  // field 1, length-delimited
+        put_varint(cr, id.size());
+        cr += id;
+    }
+    cr.push_back('\x18');  // field 3 varint (allocation_size)
+    put_varint(cr, (uint64_t)size);
+    // PreferredAllocationRequest{container_requests: [cr]}
+    std::string msg;
+    msg.push_back('\x0a');
+    put_varint(msg, cr.size());
+    msg += cr;
+    std::string framed;
+    framed.push_back('\0');
+    uint32_t be = htonl((uint32_t)msg.size());
+    framed.append((const char *)&be, 4);
+    framed += msg;
+    return framed;
+}
+
 #define NV(n, v) \
     {(uint8_t *)(n), (uint8_t *)(v), sizeof(n) - 1, sizeof(v) - 1, 0}
 
@@ -117,11 +142,25 @@ std::string allocate_request_bytes(const std::string &device_id) {
 
 int main(int argc, char **argv) {
     if (argc < 3) {
-        fprintf(stderr, "usage: %s <socket> <device-id> [iters]\n", argv[0]);
+        fprintf(stderr,
+                "usage: %s <socket> <device-id> [iters] "
+                "[pref_ids_csv] [pref_size]\n", argv[0]);
         return 2;
     }
     std::string sock_path = argv[1], device_id = argv[2];
     int iters = argc > 3 ? atoi(argv[3]) : 2000;
+    std::vector<std::string> pref_ids;
+    if (argc > 4) {
+        std::string csv = argv[4];
+        size_t pos = 0;
+        while (pos <= csv.size()) {
+            size_t c = csv.find(',', pos);
+            if (c == std::string::npos) c = csv.size();
+            if (c > pos) pref_ids.push_back(csv.substr(pos, c - pos));
+            pos = c + 1;
+        }
+    }
+    int pref_size = argc > 5 ? atoi(argv[5]) : 1;
 
     Ctx ctx;
     ctx.fd = ::socket(AF_UNIX, SOCK_STREAM, 0);
@@ -142,11 +181,9 @@ int main(int argc, char **argv) {
     ng.session_callbacks_del(cbs);
     ng.submit_settings(ctx.session, NGHTTP2_FLAG_NONE, nullptr, 0);
 
-    std::string req = allocate_request_bytes(device_id);
-    std::vector<double> lat_us;
-    lat_us.reserve(iters);
+    std::string alloc_req = allocate_request_bytes(device_id);
 
-    auto one_call = [&]() {
+    auto one_call = [&](const std::string &req, const char *path) {
         ctx.req_body = req;
         ctx.req_off = 0;
         ctx.stream_done = false;
@@ -154,7 +191,7 @@ int main(int argc, char **argv) {
         nghttp2_nv hdrs[] = {
             NV(":method", "POST"),
             NV(":scheme", "http"),
-            NV(":path", "/v1beta1.DevicePlugin/Allocate"),
+            {(uint8_t *)":path", (uint8_t *)path, 5, strlen(path), 0},
             NV(":authority", "localhost"),
             NV("content-type", "application/grpc"),
             NV("te", "trailers"),
@@ -169,20 +206,44 @@ int main(int argc, char **argv) {
         if (ctx.resp.size() < 6) { fprintf(stderr, "short response\n"); exit(1); }
     };
 
-    for (int i = 0; i < 200; ++i) one_call();  // warmup
-    for (int i = 0; i < iters; ++i) {
-        auto t0 = std::chrono::steady_clock::now();
-        one_call();
-        auto t1 = std::chrono::steady_clock::now();
-        lat_us.push_back(
-            std::chrono::duration<double, std::micro>(t1 - t0).count());
+    auto measure = [&](const std::string &req, const char *path,
+                       std::vector<double> &lat_us) {
+        for (int i = 0; i < 200; ++i) one_call(req, path);  // warmup
+        for (int i = 0; i < iters; ++i) {
+            auto t0 = std::chrono::steady_clock::now();
+            one_call(req, path);
+            auto t1 = std::chrono::steady_clock::now();
+            lat_us.push_back(
+                std::chrono::duration<double, std::micro>(t1 - t0).count());
+        }
+        std::sort(lat_us.begin(), lat_us.end());
+    };
+
+    std::vector<double> lat_us;
+    lat_us.reserve(iters);
+    measure(alloc_req, "/v1beta1.DevicePlugin/Allocate", lat_us);
+
+    std::vector<double> pref_us;
+    if (!pref_ids.empty()) {
+        std::string pref_req = preferred_request_bytes(pref_ids, pref_size);
+        pref_us.reserve(iters);
+        measure(pref_req, "/v1beta1.DevicePlugin/GetPreferredAllocation",
+                pref_us);
     }
-    std::sort(lat_us.begin(), lat_us.end());
+
     printf("{\"client\": \"native-nghttp2\", \"iters\": %d, "
-           "\"allocate_p50_us\": %.1f, \"p90_us\": %.1f, \"p99_us\": %.1f}\n",
+           "\"allocate_p50_us\": %.1f, \"p90_us\": %.1f, \"p99_us\": %.1f",
            iters, lat_us[lat_us.size() / 2],
            lat_us[(size_t)(lat_us.size() * 0.9)],
            lat_us[(size_t)(lat_us.size() * 0.99)]);
+    if (!pref_us.empty()) {
+        printf(", \"preferred_p50_us\": %.1f, \"preferred_p99_us\": %.1f, "
+               "\"preferred_size\": %d, \"preferred_pool\": %zu",
+               pref_us[pref_us.size() / 2],
+               pref_us[(size_t)(pref_us.size() * 0.99)], pref_size,
+               pref_ids.size());
+    }
+    printf("}\n");
     ng.session_del(ctx.session);
     ::close(ctx.fd);
     return 0;
