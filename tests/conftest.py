@@ -18,6 +18,7 @@ def pytest_configure(config):
 
     nb.build_drmctl()
     nb.build_fastserver()
+    nb.build_h2tool()
     try:
         nb.build_healthprobe()
     except Exception:
