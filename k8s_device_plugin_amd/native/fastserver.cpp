@@ -129,10 +129,48 @@ struct AllocState {
     std::unordered_map<uint64_t, int> weights;  // (min<<32|max) -> weight
     bool ready = false;
 
+    // Uniform group-pair weight table: the reference's weight model
+    // derives a pair's score from per-GPU facts (same devID, link type,
+    // NUMA — device.go:136-158), so every partition pair of the same two
+    // physical GPUs scores identically.  When that uniformity actually
+    // holds in the loaded topology (verified below, not assumed), the
+    // search can score candidates from per-group counts in O(G) per
+    // extension instead of O(|subset|) — exact, same totals, same
+    // tie-break order.  Any non-uniform pair (e.g. a partition missing a
+    // link entry while its siblings have one) disables the fast path.
+    bool uniform = false;
+    std::vector<std::vector<long>> gw;       // [group][group] pair weight
+    std::unordered_map<int, int> group_of_node;
+
     int weight(int a, int b) const {
         if (a > b) std::swap(a, b);
         auto it = weights.find(((uint64_t)(uint32_t)a << 32) | (uint32_t)b);
         return it == weights.end() ? 0 : it->second;
+    }
+
+    void build_uniform_table() {
+        size_t G = groups.size();
+        group_of_node.clear();
+        for (size_t g = 0; g < G; ++g)
+            for (int n : groups[g].second) group_of_node[n] = (int)g;
+        gw.assign(G, std::vector<long>(G, 0));
+        uniform = true;
+        for (size_t a = 0; a < G && uniform; ++a) {
+            for (size_t b = a; b < G && uniform; ++b) {
+                bool first = true;
+                long w0 = 0;
+                for (int i : groups[a].second) {
+                    for (int j : groups[b].second) {
+                        if (a == b && i >= j) continue;
+                        long w = weight(i, j);
+                        if (first) { w0 = w; first = false; }
+                        else if (w != w0) { uniform = false; break; }
+                    }
+                    if (!uniform) break;
+                }
+                gw[a][b] = gw[b][a] = w0;
+            }
+        }
     }
 };
 
@@ -140,6 +178,8 @@ struct Subset {
     std::vector<int> ids;
     uint64_t parents = 0;  // bitset over filtered-group indices (<=64 groups)
     long weight = 0;
+    // fast path only: per-ORIGINAL-group selected counts
+    std::vector<uint16_t> cnt;
 };
 
 bool preferred_alloc(const AllocState &st,
@@ -191,6 +231,130 @@ bool preferred_alloc(const AllocState &st,
 
     int new_size = size - (int)req_node_list.size();
 
+    const size_t n_orig = st.groups.size();
+    if (st.uniform && n_orig <= 64) {
+        // ---- closed-form fast path (exact under verified uniformity) ----
+        //
+        // BFS invariant: every group added to an incomplete state is
+        // consumed FULLY (the per-group loop only stops early when the
+        // request completes), so an intermediate state is fully described
+        // by (parent set, group add order, total size, weight).  Adding m
+        // nodes of group b to a state with per-group counts c[] costs
+        //   m * sum_h c[h]*gw[b][h]  +  C(m,2)*gw[b][b]
+        // — O(G) per GROUP instead of O(|subset|) per NODE, with no heap
+        // per state.  Candidate id lists are reconstructed only for the
+        // single winner, preserving the exact generation/insertion order
+        // of the general path (seed group ascending, then groups in added
+        // order, last group as a prefix, required ids appended last).
+        struct FS {
+            uint64_t parents = 0;
+            long weight = 0;
+            int size = 0;
+            uint8_t norder = 0;
+            uint8_t order[64];            // filtered group idx, add order
+            long rowsum[64];              // sum_h c[h]*gw[b][h] per ORIG b
+        };
+        // map filtered idx -> original group idx
+        std::vector<int> orig_of(groups.size());
+        {
+            std::unordered_map<std::string, int> orig_idx;
+            for (size_t i = 0; i < st.groups.size(); ++i)
+                orig_idx[st.groups[i].first] = (int)i;
+            // parent key may repeat only if dev_ids collide — they don't
+            for (size_t i = 0; i < groups.size(); ++i) {
+                // find by first member node (parent keys can be "")
+                orig_of[i] = st.group_of_node.at(groups[i].second[0]);
+            }
+        }
+        auto add_group = [&](FS &s, size_t fidx, int m) {
+            int b = orig_of[fidx];
+            s.weight += (long)m * s.rowsum[b] +
+                        (long)m * (m - 1) / 2 * st.gw[b][b];
+            for (size_t h = 0; h < n_orig; ++h)
+                s.rowsum[h] += (long)m * st.gw[b][h];
+            s.parents |= 1ull << fidx;
+            s.order[s.norder++] = (uint8_t)fidx;
+            s.size += m;
+        };
+        auto add_required = [&](FS &s) {
+            for (int rn : req_node_list) {
+                int b = st.group_of_node.at(rn);
+                s.weight += s.rowsum[b];
+                for (size_t h = 0; h < n_orig; ++h)
+                    s.rowsum[h] += st.gw[b][h];
+            }
+        };
+
+        std::vector<FS> queue;
+        queue.reserve(groups.size() <= 16 ? (1u << groups.size())
+                                          : 4096);
+        // parent-set dedup: direct bitmap when 2^G fits, else a set
+        std::vector<bool> seen_bm;
+        std::set<uint64_t> seen_set;
+        const bool use_bm = groups.size() <= 20;
+        if (use_bm) seen_bm.assign(1u << groups.size(), false);
+        auto seen_test_set = [&](uint64_t p) {
+            if (use_bm) {
+                if (seen_bm[p]) return true;
+                seen_bm[p] = true;
+                return false;
+            }
+            return !seen_set.insert(p).second;
+        };
+        // track the best final on the fly (first strict minimum, same
+        // tie-break as collecting then scanning) — avoids storing finals
+        FS best{};
+        bool have_best = false;
+        auto emit_final = [&](FS s) {
+            add_required(s);
+            if (!have_best || s.weight < best.weight) {
+                best = s;
+                have_best = true;
+            }
+        };
+        for (size_t idx = 0; idx < groups.size(); ++idx) {
+            FS s{};
+            int take = std::min((int)groups[idx].second.size(), new_size);
+            add_group(s, idx, take);
+            if (s.size == new_size) emit_final(s);
+            else { seen_test_set(s.parents); queue.push_back(s); }
+        }
+        for (size_t qi = 0; qi < queue.size(); ++qi) {
+            FS cur = queue[qi];
+            if (__builtin_popcountll(cur.parents) == (int)groups.size())
+                continue;
+            for (size_t idx = 0; idx < groups.size(); ++idx) {
+                if (cur.parents & (1ull << idx)) continue;
+                FS s = cur;
+                int take = std::min((int)groups[idx].second.size(),
+                                    new_size - s.size);
+                add_group(s, idx, take);
+                if (s.size == new_size) emit_final(s);
+                else if (!seen_test_set(s.parents)) {
+                    queue.push_back(s);
+                }
+            }
+        }
+        if (!have_best) { err = "no candidate subset found"; return false; }
+        // reconstruct the winner's id list in generation order
+        int remaining = new_size;
+        for (int oi = 0; oi < best.norder && remaining > 0; ++oi) {
+            const auto &g = groups[best.order[oi]].second;
+            int take = std::min((int)g.size(), remaining);
+            for (int i = 0; i < take; ++i) {
+                auto it = st.id_of_node.find(g[i]);
+                if (it != st.id_of_node.end()) out.push_back(it->second);
+            }
+            remaining -= take;
+        }
+        for (int rn : req_node_list) {
+            auto it = st.id_of_node.find(rn);
+            if (it != st.id_of_node.end()) out.push_back(it->second);
+        }
+        return true;
+    }
+
+    // ---- general path (non-uniform weights): per-node extension ----
     auto extend = [&](Subset s, int nid, int parent_idx) {
         for (int other : s.ids) s.weight += st.weight(other, nid);
         s.ids.push_back(nid);
@@ -352,6 +516,7 @@ class Server {
         // (1-kfd-visible box) the policy degrades to uniform weights and
         // every pair lookup scores 0 — the search is still well-defined
         alloc_.ready = !alloc_.groups.empty();
+        alloc_.build_uniform_table();
     }
 
     void start() {
