@@ -305,6 +305,28 @@ def _gen_memory_partitioning_supported(devices, paths, topo) -> Dict[str, str]:
     return {create_label_prefix("memory-partitioning-supported"): v}
 
 
+def _gen_xgmi_hive(devices, paths, topo) -> Dict[str, str]:
+    """xGMI hive tagging (beyond the reference; BASELINE config 3).
+
+    Values are the kfd `hive_id` in hex, counted per physical GPU — on a
+    healthy 8xMI355X node one hive of 8, so schedulers/operators can
+    select full-hive nodes (`amd.com/gpu.xgmi-hive=<id>`) and spot
+    split-hive or hive-less nodes from the counter labels.  Additionally
+    emits `amd.com/gpu.xgmi-hive-count` with the number of distinct
+    hives (0 = no xGMI info).
+    """
+    counts: Dict[str, int] = {}
+    for d in _physical_gpus(devices):
+        node = topo.node_by_render_minor(d.render_d)
+        if node is None or node.hive_id == 0:
+            continue
+        key = f"{node.hive_id:x}"
+        counts[key] = counts.get(key, 0) + 1
+    labels = _create_labels("xgmi-hive", counts) if counts else {}
+    labels[create_label_prefix("xgmi-hive-count")] = str(len(counts))
+    return labels
+
+
 LABEL_GENERATORS: Dict[str, Callable] = {
     "firmware": _gen_firmware,
     "family": _gen_family,
@@ -317,6 +339,7 @@ LABEL_GENERATORS: Dict[str, Callable] = {
     "cu-count": _gen_cu_count,
     "compute-memory-partition": _gen_compute_memory_partition,
     "compute-partitioning-supported": _gen_compute_partitioning_supported,
+    "xgmi-hive": _gen_xgmi_hive,
     "memory-partitioning-supported": _gen_memory_partitioning_supported,
 }
 

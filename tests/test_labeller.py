@@ -17,7 +17,8 @@ from k8s_device_plugin_amd.testing.fake_k8s import FakeK8s
 
 
 def test_label_kind_inventory_frozen():
-    # the 12 kinds the reference supports (main.go:115-379)
+    # the 12 kinds the reference supports (main.go:115-379) plus the
+    # beyond-reference xgmi-hive tagging (BASELINE config 3)
     assert LABEL_KINDS == sorted([
         "firmware",
         "family",
@@ -31,6 +32,7 @@ def test_label_kind_inventory_frozen():
         "compute-memory-partition",
         "compute-partitioning-supported",
         "memory-partitioning-supported",
+        "xgmi-hive",
     ])
 
 
@@ -58,6 +60,26 @@ def test_generate_labels_mi355x(fake_mi355x_8):
     # family via sysfs fallback (no ioctl on a fake tree): gfx950 -> AI
     assert labels["amd.com/gpu.family"] == "AI"
     assert labels["beta.amd.com/gpu.family.AI"] == "8"
+    # one intact xGMI hive of 8 on the fake MI355X node
+    hive_hex = f"{7455128887705989632:x}"
+    assert labels["amd.com/gpu.xgmi-hive"] == hive_hex
+    assert labels[f"beta.amd.com/gpu.xgmi-hive.{hive_hex}"] == "8"
+    assert labels["amd.com/gpu.xgmi-hive-count"] == "1"
+
+
+def test_xgmi_hive_split_node(tmp_path):
+    """Two hives of 4 -> counter labels only, hive-count=2."""
+    from k8s_device_plugin_amd.testing.fakesysfs import FakeSysfs
+
+    fs = FakeSysfs(str(tmp_path / "r"))
+    fs.add_cpu_node(0)
+    for i in range(8):
+        fs.add_physical_gpu(i, node_id=1 + i, hive_id=100 + (i // 4))
+    labels = generate_labels({"xgmi-hive": True}, fs.paths)
+    assert labels["amd.com/gpu.xgmi-hive-count"] == "2"
+    assert labels["amd.com/gpu.xgmi-hive.64"] == "4"
+    assert labels["amd.com/gpu.xgmi-hive.65"] == "4"
+    assert "amd.com/gpu.xgmi-hive" not in labels
 
 
 def test_generate_labels_subset(fake_mi355x_8):
