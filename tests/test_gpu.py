@@ -213,6 +213,12 @@ def test_labeller_on_live_sysfs(live_devices):
     assert labels["amd.com/gpu.vram"] == f"{g}G"
     assert labels["amd.com/gpu.cu-count"] == str(node.cu_count)
     assert labels["amd.com/gpu.simd-count"] == str(node.simd_count)
+    # xGMI hive tagging from the real kfd hive_id
+    if node.hive_id:
+        assert int(labels["amd.com/gpu.xgmi-hive-count"]) >= 1
+        assert f"{node.hive_id:x}" in "".join(
+            k for k in labels if "xgmi-hive" in k
+        ) or labels.get("amd.com/gpu.xgmi-hive") == f"{node.hive_id:x}"
     if backed.compute_partition == "spx":
         assert labels["amd.com/gpu.vram"] == "288G"
         assert labels["amd.com/gpu.cu-count"] == "256"
