@@ -64,6 +64,9 @@ class BestEffortPolicy:
         self._weights: Dict[int, Dict[int, int]] = {}
         self._groups: Dict[str, _Group] = {}
         self._initialized = False
+        self._uniform = False
+        self._gw: Dict[str, Dict[str, int]] = {}
+        self._group_of_node: Dict[int, str] = {}
 
     def init(
         self,
@@ -100,7 +103,53 @@ class BestEffortPolicy:
                 g.parent_id = d.id
         for g in self._groups.values():
             g.node_ids.sort()
+        self._build_uniform_table()
         self._initialized = True
+
+    def _build_uniform_table(self) -> None:
+        """Group-pair weight table when the topology is uniform.
+
+        The weight model derives a pair's score from per-GPU facts (same
+        devID, link type, NUMA, hive — weights.py), so every partition
+        pair of the same two physical GPUs normally scores identically.
+        When that holds for the loaded topology (verified here, never
+        assumed), candidate scoring collapses to per-group counts —
+        O(G) per group addition instead of O(|subset|) per node — with
+        byte-identical results (same totals, same generation order, same
+        first-minimum tie-break).  Mirrors the native server's fast path
+        (native/fastserver.cpp AllocState::build_uniform_table).
+        """
+        self._uniform = True
+        self._gw: Dict[str, Dict[str, int]] = {}
+        self._group_of_node: Dict[int, str] = {}
+        keys = list(self._groups)
+        for k in keys:
+            for n in self._groups[k].node_ids:
+                self._group_of_node[n] = k
+
+        def w(a: int, b: int) -> int:
+            if a > b:
+                a, b = b, a
+            return self._weights.get(a, {}).get(b, 0)
+
+        for i, ka in enumerate(keys):
+            ga = self._groups[ka].node_ids
+            for kb in keys[i:]:
+                gb = self._groups[kb].node_ids
+                first = True
+                w0 = 0
+                for x in ga:
+                    for y in gb:
+                        if ka == kb and x >= y:
+                            continue
+                        wxy = w(x, y)
+                        if first:
+                            w0, first = wxy, False
+                        elif wxy != w0:
+                            self._uniform = False
+                            return
+                self._gw.setdefault(ka, {})[kb] = w0
+                self._gw.setdefault(kb, {})[ka] = w0
 
     @property
     def initialized(self) -> bool:
@@ -142,6 +191,8 @@ class BestEffortPolicy:
 
         available = [self._devices[i] for i in available_ids if i in self._devices]
         required = [self._devices[i] for i in required_ids if i in self._devices]
+        if self._uniform:
+            return self._allocate_uniform(available, required, size)
         candidates = self._candidate_subsets(available, required, size)
         if not candidates:
             raise AllocationError("no candidate subset found with matching criteria")
@@ -149,6 +200,92 @@ class BestEffortPolicy:
         best = min(candidates, key=lambda s: s.weight)
         by_node = {d.node_id: d.id for d in available}
         return [by_node[nid] for nid in best.ids if nid in by_node]
+
+    def _allocate_uniform(self, available, required, size) -> List[str]:
+        """Closed-form search under verified-uniform group-pair weights.
+
+        BFS invariant: an incomplete state has consumed every added group
+        FULLY, so a state is (parent set, add order, size, weight,
+        per-group rowsums); adding m nodes of group b costs
+        m*rowsum[b] + C(m,2)*gw[b][b].  Enumeration order, totals, and
+        the first-strict-minimum tie-break match _candidate_subsets
+        exactly (cross-checked by the oracle and differential-fuzz
+        suites); the winner's id list is reconstructed in generation
+        order with required ids appended last, as the generic path does.
+        """
+        groups = self._filtered_groups(available, required)
+        new_size = size - len(required)
+        req_nodes = [d.node_id for d in required]
+        gkeys = [g.dev_id for g in groups]
+        gw = self._gw
+
+        def add_group(st, fidx, m):
+            parents, order, sz, weight, rowsum = st
+            b = gkeys[fidx]
+            weight += m * rowsum.get(b, 0) + m * (m - 1) // 2 * gw[b][b]
+            rowsum = dict(rowsum)
+            for h, wbh in gw[b].items():
+                rowsum[h] = rowsum.get(h, 0) + m * wbh
+            return (parents | (1 << fidx), order + (fidx,), sz + m,
+                    weight, rowsum)
+
+        def finish_weight(st):
+            _, _, _, weight, rowsum = st
+            rowsum = dict(rowsum)
+            for rn in req_nodes:
+                b = self._group_of_node[rn]
+                weight += rowsum.get(b, 0)
+                for h, wbh in gw[b].items():
+                    rowsum[h] = rowsum.get(h, 0) + wbh
+            return weight
+
+        best = None  # (weight, order)
+        queue = []
+        seen = set()
+        for idx, g in enumerate(groups):
+            take = min(len(g.node_ids), new_size)
+            st = add_group((0, (), 0, 0, {}), idx, take)
+            if st[2] == new_size:
+                w = finish_weight(st)
+                if best is None or w < best[0]:
+                    best = (w, st[1])
+            else:
+                seen.add(st[0])
+                queue.append(st)
+        qi = 0
+        n_groups = len(groups)
+        while qi < len(queue):
+            cur = queue[qi]
+            qi += 1
+            if bin(cur[0]).count("1") == n_groups:
+                continue
+            for idx in range(n_groups):
+                if cur[0] & (1 << idx):
+                    continue
+                take = min(len(groups[idx].node_ids), new_size - cur[2])
+                st = add_group(cur, idx, take)
+                if st[2] == new_size:
+                    w = finish_weight(st)
+                    if best is None or w < best[0]:
+                        best = (w, st[1])
+                elif st[0] not in seen:
+                    seen.add(st[0])
+                    queue.append(st)
+        if best is None:
+            raise AllocationError("no candidate subset found with matching criteria")
+
+        by_node = {d.node_id: d.id for d in available}
+        out: List[str] = []
+        remaining = new_size
+        for fidx in best[1]:
+            if remaining <= 0:
+                break
+            ids = groups[fidx].node_ids
+            take = min(len(ids), remaining)
+            out.extend(by_node[n] for n in ids[:take] if n in by_node)
+            remaining -= take
+        out.extend(by_node[n] for n in req_nodes if n in by_node)
+        return out
 
     # ---- internals ----
 
