@@ -309,3 +309,93 @@ def test_hive_and_partition_packing_combined(tmp_path):
     hives_used = {topo.nodes[devices[i].node_id].hive_id for i in out}
     assert len(hives_used) == 1, f"request straddled hives: {out}"
     assert len({devices[i].dev_id for i in out}) == 2
+
+
+def test_non_uniform_topology_uses_generic_path(tmp_path):
+    """A partition missing some inter-GPU links (weight 0 among scored
+    siblings) breaks group-pair uniformity: the closed-form fast path
+    must disable itself and the generic per-node search must serve —
+    in BOTH implementations, with identical results."""
+    import glob
+    import os
+    import shutil
+
+    import grpc
+
+    from k8s_device_plugin_amd.plugin import AMDGPUPlugin
+    from k8s_device_plugin_amd.plugin.native_server import NativePluginServer
+    from k8s_device_plugin_amd.protos import deviceplugin as dp
+
+    fs = build_mi355x_node(str(tmp_path / "r"), n_gpus=4,
+                           partitions_per_gpu=4, compute_partition="CPX")
+    # sever BOTH directions of a few specific partition pairs: those
+    # pairs now score 0 while their same-GPU-pair siblings score
+    # normally -> group-pair weights are no longer uniform
+    nodes_dir = fs.paths.kfd_topology_nodes
+    severed = {(2, 7), (2, 8), (3, 9)}
+
+    def _parse(path):
+        out = {}
+        for line in open(path):
+            k, _, v = line.partition(" ")
+            out[k.strip()] = int(v)
+        return out
+
+    removed = 0
+    for link_props in glob.glob(
+        os.path.join(nodes_dir, "*", "io_links", "*", "properties")
+    ):
+        p = _parse(link_props)
+        pair = tuple(sorted((p.get("node_from", -1), p.get("node_to", -1))))
+        if pair in severed:
+            shutil.rmtree(os.path.dirname(link_props))
+            removed += 1
+    assert removed == 2 * len(severed), removed
+
+    topo = KFDTopology.load(fs.paths)
+    devices = discover_gpus(fs.paths, topology=topo)
+    policy = BestEffortPolicy()
+    policy.init(devices.values(), topology=topo)
+    assert not policy._uniform, "fast path must disable on non-uniform weights"
+
+    ids = sorted(devices)
+    plugin = AMDGPUPlugin(resource="gpu", paths=fs.paths)
+    plugin.start()
+    srv = NativePluginServer(plugin, str(tmp_path / "s.sock"))
+    srv.start()
+    try:
+        ch = grpc.insecure_channel(f"unix://{tmp_path}/s.sock")
+        stub = dp.DevicePluginStub(ch)
+        import random
+
+        rng = random.Random(5)
+        for size in list(range(1, 16)) + [16]:
+            py = policy.allocate(ids, [], size)
+            req = dp.PreferredAllocationRequest()
+            cr = req.container_requests.add()
+            cr.available_deviceIDs.extend(ids)
+            cr.allocation_size = size
+            native = list(
+                stub.GetPreferredAllocation(req, timeout=30)
+                .container_responses[0].deviceIDs
+            )
+            assert native == py, (size, native, py)
+        for _ in range(40):
+            av = rng.sample(ids, rng.randint(2, len(ids)))
+            size = rng.randint(1, len(av))
+            required = rng.sample(av, rng.randint(0, min(2, size)))
+            py = policy.allocate(av, required, size)
+            req = dp.PreferredAllocationRequest()
+            cr = req.container_requests.add()
+            cr.available_deviceIDs.extend(av)
+            cr.must_include_deviceIDs.extend(required)
+            cr.allocation_size = size
+            native = list(
+                stub.GetPreferredAllocation(req, timeout=30)
+                .container_responses[0].deviceIDs
+            )
+            assert native == py, (size, av, required)
+        ch.close()
+    finally:
+        srv.stop()
+        plugin.stop()
