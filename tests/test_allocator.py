@@ -399,3 +399,37 @@ def test_non_uniform_topology_uses_generic_path(tmp_path):
     finally:
         srv.stop()
         plugin.stop()
+
+
+def test_uniform_fast_path_property(tmp_path):
+    """Property: on every uniform topology shape, the closed-form fast
+    path and the generic per-node search return IDENTICAL lists for all
+    full-pool sizes and a seeded sample of partial requests."""
+    import random
+
+    rng = random.Random(0xFA57)
+    for n_gpus, parts in ((2, 1), (3, 2), (4, 4), (8, 1), (4, 8)):
+        fs = build_mi355x_node(
+            str(tmp_path / f"u{n_gpus}x{parts}"), n_gpus=n_gpus,
+            partitions_per_gpu=parts,
+            compute_partition="CPX" if parts > 1 else "SPX",
+        )
+        topo = KFDTopology.load(fs.paths)
+        devices = discover_gpus(fs.paths, topology=topo)
+        fast = BestEffortPolicy()
+        fast.init(devices.values(), topology=topo)
+        assert fast._uniform
+        generic = BestEffortPolicy()
+        generic.init(devices.values(), topology=topo)
+        generic._uniform = False
+
+        ids = sorted(devices)
+        for size in range(1, len(ids) + 1):
+            assert fast.allocate(ids, [], size) == \
+                generic.allocate(ids, [], size), (n_gpus, parts, size)
+        for _ in range(25):
+            av = rng.sample(ids, rng.randint(2, len(ids)))
+            size = rng.randint(1, len(av))
+            req = rng.sample(av, rng.randint(0, min(2, size)))
+            assert fast.allocate(av, req, size) == \
+                generic.allocate(av, req, size), (n_gpus, parts, size, av, req)
