@@ -178,8 +178,6 @@ struct Subset {
     std::vector<int> ids;
     uint64_t parents = 0;  // bitset over filtered-group indices (<=64 groups)
     long weight = 0;
-    // fast path only: per-ORIGINAL-group selected counts
-    std::vector<uint16_t> cnt;
 };
 
 bool preferred_alloc(const AllocState &st,
