@@ -34,6 +34,7 @@ def main() -> int:
     ap.add_argument("--deep-every", type=int, default=3)
     ap.add_argument("--metrics-port", type=int, default=19793)
     ap.add_argument("--kubelet-restart-s", type=float, default=240.0)
+    ap.add_argument("--server", default="native", choices=["native", "python"])
     args = ap.parse_args()
 
     from k8s_device_plugin_amd.protos import deviceplugin as dp
@@ -61,6 +62,7 @@ def main() -> int:
          "--sysroot", sysroot, "--kubelet-dir", dp_dir,
          "-pulse", str(args.pulse),
          "--deep-probe-every", str(args.deep_every),
+         "--server", args.server,
          "--cdi", "--cdi-dir", cdi_dir,
          "--metrics-port", str(args.metrics_port), "-v"],
         stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
@@ -196,6 +198,7 @@ def main() -> int:
     )
     out = {
         "live_sysfs": live,
+        "server_impl": args.server,
         "minutes": args.minutes,
         "pulse_s": args.pulse,
         "deep_probe_every": args.deep_every,
