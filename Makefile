@@ -39,3 +39,11 @@ helm:
 clean:
 	rm -f k8s_device_plugin_amd/native/*.so
 	find . -name __pycache__ -type d -exec rm -rf {} +
+
+# ASAN + TSAN over the native-server matrix; logs in profiles/sanitizers/
+sanitize:
+	bash scripts/run_sanitizers.sh all
+
+# 10-minute live-daemon soak (native server; use SOAK_ARGS to customize)
+soak:
+	python3 scripts/soak.py --minutes 10 --pulse 2 --deep-every 3 $(SOAK_ARGS)
