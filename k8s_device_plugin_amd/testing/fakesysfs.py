@@ -278,6 +278,11 @@ def build_mi355x_node(
     next_minor = 128 + n_gpus  # partition renderDs start after the physical ones
 
     for i in range(n_gpus):
+        # in CPX-style trees the physical GPU keeps the FIRST kfd node,
+        # and that node reports per-partition values like every other
+        # partition node (real MI300/MI355 CPX behavior; cf. the
+        # topo-mi300-cpx reference fixture where all 8 nodes of a GPU
+        # carry the divided simd counts)
         fs.add_physical_gpu(
             i,
             node_id=next_node,
@@ -285,6 +290,8 @@ def build_mi355x_node(
             memory_partition=memory_partition,
             numa_node=numa_map[i],
             hive_id=hive_id,
+            vram_bytes=MI355X_VRAM_BYTES // partitions_per_gpu,
+            simd_count=MI355X_SIMD_COUNT // partitions_per_gpu,
         )
         gpu_nodes.append(next_node)
         next_node += 1

@@ -182,3 +182,34 @@ def test_cli_show_and_gated_set(fs, monkeypatch, capsys):
     assert all(
         v[0] == "CPX" for v in current_partition_modes(fs.paths).values()
     )
+
+
+def test_mode_flip_updates_labels(tmp_path, allow_env):
+    """Partition flip + labeller refresh: after SPX->CPX the recomputed
+    labels must advertise the new partition config and per-partition
+    values (the labeller's --refresh-interval path picks this up without
+    a restart; the ROCm labeller needs a pod restart)."""
+    import shutil
+
+    from k8s_device_plugin_amd.labeller import generate_labels
+    from k8s_device_plugin_amd.labeller.labels import LABEL_KINDS
+
+    root = str(tmp_path / "node")
+    build_mi355x_node(root, n_gpus=2)
+    paths = SysPaths(root)
+    enabled = {k: True for k in LABEL_KINDS}
+
+    before = generate_labels(enabled, paths)
+    assert before["amd.com/gpu.compute-memory-partition"] == "spx_nps1"
+    assert before["amd.com/gpu.cu-count"] == "256"
+
+    set_partition_mode(paths, compute="CPX", memory="NPS2", allow=True)
+    shutil.rmtree(root)
+    build_mi355x_node(root, n_gpus=2, partitions_per_gpu=8,
+                      compute_partition="CPX", memory_partition="NPS2")
+
+    after = generate_labels(enabled, paths)
+    assert after["amd.com/gpu.compute-memory-partition"] == "cpx_nps2"
+    # per-partition CU count: 256/8 = 32 per logical device
+    assert after["amd.com/gpu.cu-count"] == "32"
+    assert after["amd.com/gpu.vram"] == "36G"  # 288G / 8 partitions
