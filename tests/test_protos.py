@@ -77,3 +77,32 @@ def test_constants_match_kubelet_abi():
     assert dp.KUBELET_SOCKET == "/var/lib/kubelet/device-plugins/kubelet.sock"
     assert dp.HEALTHY == "Healthy" and dp.UNHEALTHY == "Unhealthy"
     assert ms.EXPORTER_SOCKET.endswith("amdgpu_device_metrics_exporter_grpc.socket")
+
+
+def test_podresources_golden_bytes():
+    """kubelet PodResources v1 ABI pins (field numbers are the wire
+    contract; see protos/podresources.py)."""
+    from k8s_device_plugin_amd.protos import podresources as pr
+
+    cd = pr.ContainerDevices(resource_name="amd.com/gpu",
+                             device_ids=["x", "y"])
+    expect = (
+        b"\x0a\x0b" + b"amd.com/gpu"   # resource_name field 1
+        + b"\x12\x01x" + b"\x12\x01y"  # device_ids field 2
+    )
+    assert cd.SerializeToString() == expect
+
+    req = pr.GetPodResourcesRequest(pod_name="p", pod_namespace="ns")
+    assert req.SerializeToString() == b"\x0a\x01p\x12\x02ns"
+
+    pod = pr.PodResources(name="p", namespace="ns")
+    c = pod.containers.add()
+    c.name = "main"
+    c.cpu_ids.extend([3, 4])
+    inner = b"\x0a\x04main" + b"\x1a\x02\x03\x04"  # name f1, cpu_ids f3 packed
+    assert pod.SerializeToString() == (
+        b"\x0a\x01p\x12\x02ns" + b"\x1a" + bytes([len(inner)]) + inner
+    )
+
+    assert pr.POD_RESOURCES_SERVICE == "v1.PodResourcesLister"
+    assert pr.PODRESOURCES_SOCKET == "/var/lib/kubelet/pod-resources/kubelet.sock"
