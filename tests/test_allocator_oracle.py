@@ -250,3 +250,57 @@ def test_mi300cpx_group_shapes(mi300cpx):
     policy, _, _ = mi300cpx
     sizes = sorted(len(g.node_ids) for g in policy._groups.values())
     assert sizes == [7, 8, 8, 8, 8, 8, 8, 8]
+
+
+# ---- the same frozen oracles through the NATIVE server ----
+
+@pytest.fixture(scope="module")
+def native_oracle_server(mi300cpx, tmp_path_factory):
+    """A bare _fastserver.Server loaded with the mi300-cpx oracle policy
+    state — the reference's hardest frozen cases answered by the C++
+    search (closed-form fast path on this uniform topology)."""
+    from k8s_device_plugin_amd.native import load_fastserver
+
+    mod = load_fastserver()
+    if mod is None:
+        pytest.skip("fastserver extension unavailable")
+    policy, devices, _ = mi300cpx
+    sock = str(tmp_path_factory.mktemp("native_oracle") / "s.sock")
+    srv = mod.Server(sock)
+    srv.set_list_response(b"")
+    srv.set_allocator_state(*policy.export_state())
+    srv.start()
+    yield devices, sock
+    srv.stop()
+
+
+@pytest.mark.parametrize("desc,available,filtered,required,size,expected",
+                         MI300_CASES, ids=["native:" + c[0] for c in MI300_CASES])
+def test_mi300cpx_oracle_native(native_oracle_server, desc, available,
+                                filtered, required, size, expected):
+    import grpc
+
+    from k8s_device_plugin_amd.protos import deviceplugin as dp
+
+    devices, sock = native_oracle_server
+    if available:
+        av = list(available)
+    else:
+        av = [d.id for d in devices]
+    if filtered:
+        av = [a for a in av if a not in set(filtered)]
+    ch = grpc.insecure_channel(f"unix://{sock}")
+    stub = dp.DevicePluginStub(ch)
+    req = dp.PreferredAllocationRequest()
+    cr = req.container_requests.add()
+    cr.available_deviceIDs.extend(av)
+    cr.must_include_deviceIDs.extend(required)
+    cr.allocation_size = size
+    out = list(
+        stub.GetPreferredAllocation(req, timeout=30)
+        .container_responses[0].deviceIDs
+    )
+    ch.close()
+    assert len(out) == size
+    if expected is not None:
+        assert sorted(out) == sorted(expected), f"{desc}: got {sorted(out)}"
