@@ -304,3 +304,44 @@ def test_mi300cpx_oracle_native(native_oracle_server, desc, available,
     assert len(out) == size
     if expected is not None:
         assert sorted(out) == sorted(expected), f"{desc}: got {sorted(out)}"
+
+
+@pytest.mark.parametrize("desc,available,required,size,expected",
+                         MI210_CASES, ids=["native:" + c[0] for c in MI210_CASES])
+def test_mi210_oracle_native(mi210, tmp_path, desc, available, required,
+                             size, expected):
+    """The MI210 whole-GPU oracles through the C++ search as well."""
+    import grpc
+
+    from k8s_device_plugin_amd.native import load_fastserver
+    from k8s_device_plugin_amd.protos import deviceplugin as dp
+
+    mod = load_fastserver()
+    if mod is None:
+        pytest.skip("fastserver extension unavailable")
+    policy, devices, _ = mi210
+    sock = str(tmp_path / "s.sock")
+    srv = mod.Server(sock)
+    srv.set_list_response(b"")
+    srv.set_allocator_state(*policy.export_state())
+    srv.start()
+    try:
+        av = available if available else sorted(
+            (d.id for d in devices), key=lambda s: next(
+                d.node_id for d in devices if d.id == s)
+        )
+        ch = grpc.insecure_channel(f"unix://{sock}")
+        stub = dp.DevicePluginStub(ch)
+        req = dp.PreferredAllocationRequest()
+        cr = req.container_requests.add()
+        cr.available_deviceIDs.extend(av)
+        cr.must_include_deviceIDs.extend(required)
+        cr.allocation_size = size
+        out = list(
+            stub.GetPreferredAllocation(req, timeout=30)
+            .container_responses[0].deviceIDs
+        )
+        ch.close()
+        assert sorted(out) == sorted(expected), f"{desc}: got {sorted(out)}"
+    finally:
+        srv.stop()

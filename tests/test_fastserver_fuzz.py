@@ -1,11 +1,15 @@
 """Differential fuzz: the native C++ preferred-allocation search must agree
 with the Python policy on random topologies and random requests."""
 
+import os
 import random
 
 import grpc
 import hypothesis.strategies as st
 from hypothesis import given, settings
+
+# long local runs: AMDXDP_FUZZ_EXAMPLES=300 pytest tests/test_fastserver_fuzz.py
+_EXAMPLES = int(os.environ.get("AMDXDP_FUZZ_EXAMPLES", "12"))
 
 from k8s_device_plugin_amd.plugin import AMDGPUPlugin
 from k8s_device_plugin_amd.plugin.native_server import NativePluginServer
@@ -17,7 +21,7 @@ from k8s_device_plugin_amd.protos import deviceplugin as dp
     parts=st.sampled_from([1, 2, 4]),
     seed=st.integers(min_value=0, max_value=2**31),
 )
-@settings(max_examples=12, deadline=None)
+@settings(max_examples=_EXAMPLES, deadline=None)
 def test_native_matches_python_random(tmp_path_factory, n_gpus, parts, seed):
     from k8s_device_plugin_amd.testing.fakesysfs import build_mi355x_node
 
