@@ -415,11 +415,31 @@ py::dict p2p_bandwidth(int src, int dst, size_t bytes) {
 
 } // namespace
 
+// PCI bus id per HIP ordinal ("domain:bus:device.function", lowercase) —
+// lets the plugin map its kfd-derived device ids onto ordinals exactly
+// instead of assuming enumeration order (ROCR_VISIBLE_DEVICES can
+// reorder it).
+std::vector<std::string> pci_bus_ids() {
+    int n = 0;
+    HIP_CHECK(hipGetDeviceCount(&n));
+    std::vector<std::string> out;
+    out.reserve(n);
+    for (int i = 0; i < n; ++i) {
+        char buf[32] = {0};
+        if (hipDeviceGetPCIBusId(buf, sizeof(buf), i) == hipSuccess)
+            out.emplace_back(buf);
+        else
+            out.emplace_back("");
+    }
+    return out;
+}
+
 PYBIND11_MODULE(_healthprobe, m) {
     m.doc() = "gfx950 deep GPU health probe (MFMA/LDS/HBM)";
     m.def("run_probe", &run_probe, py::arg("device") = 0,
           py::arg("hbm_bytes") = (size_t)1 << 30);
     m.def("device_count", &device_count);
+    m.def("pci_bus_ids", &pci_bus_ids);
     m.def("p2p_bandwidth", &p2p_bandwidth, py::arg("src") = 0,
           py::arg("dst") = 1, py::arg("bytes") = (size_t)1 << 30);
     m.def("mfma_probe_raw", &mfma_probe_raw, py::arg("device") = 0);

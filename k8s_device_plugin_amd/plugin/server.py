@@ -166,10 +166,25 @@ class AMDGPUPlugin:
             self._run_deep_check()
 
     def _hip_ordinal(self, dev: GPUDevice) -> int:
-        """Best-effort HIP device ordinal for a plugin device: HIP
-        enumerates GPUs in PCI bus order, and dev_id is the PCI-ish
-        address shared by a GPU's partitions, so the sorted physical
-        dev_id rank is the ordinal."""
+        """HIP device ordinal for a plugin device.
+
+        Primary: exact match of the device's PCI address (dev_id, format
+        "dddd:bb:dd:f") against the runtime's per-ordinal PCI bus ids
+        ("dddd:bb:dd.f") — robust against ROCR_VISIBLE_DEVICES
+        reordering.  Fallback when the probe extension can't enumerate:
+        sorted physical dev_id rank (HIP's default PCI enumeration
+        order)."""
+        try:
+            from ..native import load_healthprobe
+
+            mod = load_healthprobe(required=False)
+            if mod is not None and hasattr(mod, "pci_bus_ids"):
+                want = dev.dev_id.lower().replace(".", ":")
+                for ordinal, bus in enumerate(mod.pci_bus_ids()):
+                    if bus and bus.lower().replace(".", ":") == want:
+                        return ordinal
+        except Exception:  # enumeration impossible (no GPU) -> fallback
+            pass
         phys = sorted({d.dev_id for d in self.devices.values() if d.kfd_backed})
         return phys.index(dev.dev_id)
 

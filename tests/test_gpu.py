@@ -439,3 +439,28 @@ def test_deep_probe_heartbeat_live(live_devices):
         assert health[did] == "Healthy", (did, plugin._deep_failed)
     assert not plugin._deep_failed
     plugin.stop()
+
+
+def test_hip_ordinal_pci_mapping(live_devices):
+    """The plugin's device->HIP-ordinal mapping must go through the
+    runtime's PCI bus ids (exact match), not enumeration-order guessing."""
+    import torch
+
+    from k8s_device_plugin_amd.native import load_healthprobe
+    from k8s_device_plugin_amd.plugin import AMDGPUPlugin
+    from k8s_device_plugin_amd.topology import SysPaths
+
+    mod = load_healthprobe(required=True)
+    bus_ids = mod.pci_bus_ids()
+    assert len(bus_ids) == torch.cuda.device_count()
+    assert all(b for b in bus_ids), bus_ids
+
+    plugin = AMDGPUPlugin(resource="gpu", paths=SysPaths("/"))
+    plugin.start()
+    backed = [d for d in plugin.devices.values() if d.kfd_backed]
+    for d in backed:
+        ordinal = plugin._hip_ordinal(d)
+        assert 0 <= ordinal < len(bus_ids)
+        assert bus_ids[ordinal].lower().replace(".", ":") == \
+            d.dev_id.lower().replace(".", ":")
+    plugin.stop()
