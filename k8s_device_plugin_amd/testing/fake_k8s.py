@@ -19,6 +19,9 @@ class FakeK8s:
         self.node_name = node_name
         self.labels: Dict[str, str] = dict(initial_labels or {})
         self.patches = []  # applied label patches, in order
+        self.watch_requests = []  # raw query strings of watch GETs
+        self.resource_version = 100  # bumped on every mutation/event
+        self.fail_next_watch_410 = False  # one-shot HTTP 410 answer
         self._events: "queue.Queue" = queue.Queue()
         self._server: Optional[ThreadingHTTPServer] = None
         self._thread: Optional[threading.Thread] = None
@@ -31,11 +34,38 @@ class FakeK8s:
             return {
                 "apiVersion": "v1",
                 "kind": "Node",
-                "metadata": {"name": self.node_name, "labels": dict(self.labels)},
+                "metadata": {
+                    "name": self.node_name,
+                    "labels": dict(self.labels),
+                    "resourceVersion": str(self.resource_version),
+                },
             }
 
     def push_event(self, evt_type: str = "ADDED") -> None:
+        with self._lock:
+            self.resource_version += 1
         self._events.put({"type": evt_type, "object": self.node_object()})
+
+    def push_bookmark(self) -> None:
+        """Apiserver progress bookmark (allowWatchBookmarks=true)."""
+        with self._lock:
+            self.resource_version += 1
+        self._events.put({
+            "type": "BOOKMARK",
+            "object": {
+                "apiVersion": "v1",
+                "kind": "Node",
+                "metadata": {"name": self.node_name,
+                             "resourceVersion": str(self.resource_version)},
+            },
+        })
+
+    def push_gone_error(self) -> None:
+        """In-stream 410 Gone Status (resourceVersion too old)."""
+        self._events.put({
+            "type": "ERROR",
+            "object": {"kind": "Status", "code": 410, "reason": "Expired"},
+        })
 
     # ---- server ----
 
@@ -56,6 +86,12 @@ class FakeK8s:
 
             def do_GET(self):
                 if self.path.startswith("/api/v1/nodes?watch=true"):
+                    fake.watch_requests.append(self.path)
+                    if fake.fail_next_watch_410:
+                        fake.fail_next_watch_410 = False
+                        self._json(410, {"kind": "Status", "code": 410,
+                                         "reason": "Expired"})
+                        return
                     self.send_response(200)
                     self.send_header("Content-Type", "application/json")
                     self.end_headers()
@@ -79,6 +115,7 @@ class FakeK8s:
                 body = json.loads(self.rfile.read(length) or b"{}")
                 patch = body.get("metadata", {}).get("labels", {})
                 with fake._lock:
+                    fake.resource_version += 1
                     fake.patches.append(dict(patch))
                     for k, v in patch.items():
                         if v is None:

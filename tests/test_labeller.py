@@ -298,3 +298,98 @@ def test_update_labels_reconciles_on_change(fake_mi355x_8):
         assert fake.labels["beta.amd.com/gpu.vram.144G"] == "1"
     finally:
         fake.stop()
+
+
+def test_watch_resumes_with_resource_version():
+    """Informer semantics: reconnects resume from the last seen
+    resourceVersion (bookmarks advance it without firing on_event), and
+    a 410 Gone clears the bookmark so the next watch relists."""
+    import threading
+    import time
+
+    fake = FakeK8s("node-0").start()
+    try:
+        client = K8sClient(base_url=fake.base_url, token="t")
+        events = []
+        stop = threading.Event()
+        th = threading.Thread(
+            target=client.watch_node,
+            args=("node-0", lambda t, o: events.append((t, o)), stop),
+            kwargs={"timeout_seconds": 1},
+            daemon=True,
+        )
+        th.start()
+
+        fake.push_event("MODIFIED")       # rv bumps; delivered
+        fake.push_bookmark()              # rv bumps; NOT delivered
+        deadline = time.monotonic() + 5
+        while len(events) < 1 and time.monotonic() < deadline:
+            time.sleep(0.05)
+        assert len(events) == 1 and events[0][0] == "MODIFIED"
+        last_rv = fake.resource_version
+
+        # stream ends (1 s server timeout) -> client reconnects with the
+        # bookmark's rv
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline:
+            resumed = [
+                q for q in fake.watch_requests
+                if f"resourceVersion={last_rv}" in q
+            ]
+            if resumed:
+                break
+            time.sleep(0.1)
+        assert resumed, fake.watch_requests
+
+        # 410 on the next reconnect -> rv cleared -> relist (no rv param)
+        n_before = len(fake.watch_requests)
+        fake.fail_next_watch_410 = True
+        deadline = time.monotonic() + 10
+        cleared = False
+        while time.monotonic() < deadline:
+            newer = fake.watch_requests[n_before + 1:]
+            if any("resourceVersion" not in q for q in newer):
+                cleared = True
+                break
+            time.sleep(0.1)
+        assert cleared, fake.watch_requests[n_before:]
+        stop.set()
+        th.join(timeout=5)
+    finally:
+        fake.stop()
+
+
+def test_watch_instream_gone_clears_bookmark():
+    """An in-stream ERROR Status (410 inside the watch body) must also
+    clear the bookmark and trigger a relist."""
+    import threading
+    import time
+
+    fake = FakeK8s("node-0").start()
+    try:
+        client = K8sClient(base_url=fake.base_url, token="t")
+        stop = threading.Event()
+        th = threading.Thread(
+            target=client.watch_node,
+            args=("node-0", lambda t, o: None, stop),
+            kwargs={"timeout_seconds": 1},
+            daemon=True,
+        )
+        th.start()
+        fake.push_event("MODIFIED")   # establish a bookmark
+        time.sleep(0.3)
+        fake.push_gone_error()        # stream-level 410
+        deadline = time.monotonic() + 10
+        ok = False
+        while time.monotonic() < deadline:
+            # after the ERROR the client must reconnect WITHOUT an rv
+            tail = fake.watch_requests[1:]
+            if any("resourceVersion" not in q for q in tail):
+                ok = True
+                break
+            time.sleep(0.1)
+        assert ok, fake.watch_requests
+        stop.set()
+        th.join(timeout=5)
+    finally:
+        fake.stop()
