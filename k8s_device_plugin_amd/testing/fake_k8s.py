@@ -20,6 +20,7 @@ class FakeK8s:
         self.labels: Dict[str, str] = dict(initial_labels or {})
         self.patches = []  # applied label patches, in order
         self.watch_requests = []  # raw query strings of watch GETs
+        self.auth_headers = []  # Authorization header of every request
         self.resource_version = 100  # bumped on every mutation/event
         self.fail_next_watch_410 = False  # one-shot HTTP 410 answer
         self._events: "queue.Queue" = queue.Queue()
@@ -85,6 +86,7 @@ class FakeK8s:
                 self.wfile.write(body)
 
             def do_GET(self):
+                fake.auth_headers.append(self.headers.get("Authorization"))
                 if self.path.startswith("/api/v1/nodes?watch=true"):
                     fake.watch_requests.append(self.path)
                     if fake.fail_next_watch_410:
@@ -108,6 +110,7 @@ class FakeK8s:
                     self._json(404, {"kind": "Status", "code": 404})
 
             def do_PATCH(self):
+                fake.auth_headers.append(self.headers.get("Authorization"))
                 if self.path != f"/api/v1/nodes/{fake.node_name}":
                     self._json(404, {"kind": "Status", "code": 404})
                     return

@@ -393,3 +393,27 @@ def test_watch_instream_gone_clears_bookmark():
         th.join(timeout=5)
     finally:
         fake.stop()
+
+
+def test_sa_token_rotation_picked_up(tmp_path, monkeypatch):
+    """Bound SA tokens rotate ~hourly; every request must re-read the
+    projected token file (a pinned header 401s forever after
+    the first rotation)."""
+    import k8s_device_plugin_amd.labeller.k8s as k8smod
+
+    sa_dir = tmp_path / "sa"
+    sa_dir.mkdir()
+    (sa_dir / "token").write_text("token-v1\n")
+    monkeypatch.setattr(k8smod, "SA_DIR", str(sa_dir))
+
+    fake = FakeK8s("node-0").start()
+    try:
+        client = K8sClient(base_url=fake.base_url)  # token from file
+        client.get_node("node-0")
+        assert fake.auth_headers[-1] == "Bearer token-v1"
+        # kubelet rotates the projected file in place
+        (sa_dir / "token").write_text("token-v2\n")
+        client.patch_node_labels("node-0", {"x": "y"})
+        assert fake.auth_headers[-1] == "Bearer token-v2"
+    finally:
+        fake.stop()
