@@ -156,6 +156,8 @@ def device_plugin_main(argv=None) -> int:
                 "pair_weights": [
                     {"from": a, "to": b, "weight": w} for a, b, w in weights
                 ],
+                # closed-form fast path active? (uniform group-pair weights)
+                "uniform_fast_path": policy._uniform,
             }
         except AllocationError as e:
             out["allocator"] = {"error": str(e)}
