@@ -28,6 +28,7 @@
 
 #include <algorithm>
 #include <atomic>
+#include <chrono>
 #include <cstdint>
 #include <map>
 #include <memory>
@@ -488,6 +489,10 @@ class Server {
         d["unknown_method_total"] = (uint64_t)n_unknown_.load();
         d["list_pushes_total"] = (uint64_t)n_pushes_.load();
         d["connections_total"] = (uint64_t)n_conns_.load();
+        // handler-time sums (ns): with the *_total counters these give
+        // Prometheus-style average handler latency via rate()/rate()
+        d["allocate_handler_ns_total"] = (uint64_t)allocate_ns_.load();
+        d["preferred_handler_ns_total"] = (uint64_t)preferred_ns_.load();
         return d;
     }
 
@@ -713,11 +718,22 @@ class Server {
             resp = "";
         } else if (p == "/v1beta1.DevicePlugin/Allocate") {
             n_allocate_.fetch_add(1, std::memory_order_relaxed);
+            auto t0 = std::chrono::steady_clock::now();
             resp = handle_allocate(msg);
+            allocate_ns_.fetch_add(
+                (uint64_t)std::chrono::duration_cast<std::chrono::nanoseconds>(
+                    std::chrono::steady_clock::now() - t0).count(),
+                std::memory_order_relaxed);
         } else if (p == "/v1beta1.DevicePlugin/GetPreferredAllocation") {
             n_preferred_.fetch_add(1, std::memory_order_relaxed);
+            auto t0 = std::chrono::steady_clock::now();
             std::string err;
-            if (!handle_preferred(msg, resp, err)) {
+            bool ok = handle_preferred(msg, resp, err);
+            preferred_ns_.fetch_add(
+                (uint64_t)std::chrono::duration_cast<std::chrono::nanoseconds>(
+                    std::chrono::steady_clock::now() - t0).count(),
+                std::memory_order_relaxed);
+            if (!ok) {
                 st->grpc_status = "3";  // INVALID_ARGUMENT
                 st->grpc_message = err;
                 resp.clear();
@@ -1028,6 +1044,7 @@ class Server {
     std::atomic<bool> running_{false};
     std::vector<std::unique_ptr<Conn>> conns_;
 
+    std::atomic<uint64_t> allocate_ns_{0}, preferred_ns_{0};
     std::atomic<uint64_t> n_allocate_{0}, n_preferred_{0}, n_listwatch_{0},
         n_options_{0}, n_prestart_{0}, n_unknown_{0}, n_pushes_{0},
         n_conns_{0};

@@ -16,6 +16,8 @@ log = logging.getLogger(__name__)
 
 _COUNTER_HELP = {
     "allocate_total": "Allocate RPCs served",
+    "allocate_handler_ns_total": "cumulative Allocate handler time (ns)",
+    "preferred_handler_ns_total": "cumulative GetPreferredAllocation handler time (ns)",
     "preferred_allocation_total": "GetPreferredAllocation RPCs served",
     "list_and_watch_streams_total": "ListAndWatch streams opened",
     "options_total": "GetDevicePluginOptions RPCs served",
