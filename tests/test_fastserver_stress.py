@@ -175,3 +175,42 @@ def test_large_response_multi_frame(native):
     assert len(resp.container_responses[-1].devices) == 1 + 2 * len(ids)
     assert len(resp.SerializeToString()) > 500_000
     ch.close()
+
+
+def test_fifty_concurrent_streams(native):
+    """50 simultaneous ListAndWatch streams (one channel each, forcing
+    separate connections is not possible with grpc pooling — mix grpc
+    channels and raw go-wire connections): every stream must get the
+    initial list AND a heartbeat push; no fd/slot exhaustion."""
+    from k8s_device_plugin_amd.protos import deviceplugin as dp
+    from k8s_device_plugin_amd.testing.goclient import GoWireClient
+
+    plugin, srv, sock = native
+    # 25 raw connections, each with one ListAndWatch stream
+    raws = []
+    for _ in range(25):
+        c = GoWireClient(sock)
+        sid = c.start_call("/v1beta1.DevicePlugin/ListAndWatch", b"")
+        raws.append((c, c.conn.stream(sid)))
+    # 25 grpc streams (share a channel/connection, distinct h2 streams)
+    ch = grpc.insecure_channel(f"unix://{sock}")
+    stub = dp.DevicePluginStub(ch)
+    calls = [stub.ListAndWatch(dp.Empty()) for _ in range(25)]
+    iters = [iter(c) for c in calls]
+    for it in iters:
+        first = next(it)
+        assert len(first.devices) == 8
+    for c, st in raws:
+        assert c.conn.wait(lambda: len(st.grpc_messages()) >= 1, timeout=10)
+
+    srv.heartbeat()  # one push must reach all 50 streams
+    for it in iters:
+        assert len(next(it).devices) == 8
+    for c, st in raws:
+        assert c.conn.wait(lambda: len(st.grpc_messages()) >= 2, timeout=10)
+
+    for call in calls:
+        call.cancel()
+    ch.close()
+    for c, _ in raws:
+        c.close()
