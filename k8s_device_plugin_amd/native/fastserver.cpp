@@ -882,10 +882,19 @@ class Server {
         (void)rc;
     }
 
+    // Connection cap: the kubelet needs 1-2 connections; a runaway local
+    // client must not exhaust the daemon's fds (the socket dir is
+    // root-only, so this is belt-and-braces, not a security boundary).
+    static constexpr size_t kMaxConns = 256;
+
     void accept_conn() {
         for (;;) {
             int fd = ::accept4(listen_fd_, nullptr, nullptr, SOCK_NONBLOCK);
             if (fd < 0) return;
+            if (conns_.size() >= kMaxConns) {
+                ::close(fd);  // refuse; peer sees ECONNRESET and retries
+                continue;
+            }
             auto &ng = NgHttp2::get();
             nghttp2_session_callbacks *cbs;
             ng.session_callbacks_new(&cbs);
