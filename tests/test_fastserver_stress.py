@@ -214,3 +214,41 @@ def test_fifty_concurrent_streams(native):
     ch.close()
     for c, _ in raws:
         c.close()
+
+
+def test_connection_cap_protects_fds(native):
+    """Beyond 256 connections the server refuses new ones instead of
+    exhausting fds — and the kubelet's existing connection keeps
+    working throughout."""
+    import socket as socketmod
+
+    plugin, srv, sock = native
+    ch = grpc.insecure_channel(f"unix://{sock}")
+    stub = dp.DevicePluginStub(ch)
+    req = dp.AllocateRequest()
+    req.container_requests.add().devices_ids.append(sorted(plugin.devices)[0])
+    stub.Allocate(req, timeout=10)  # kubelet connection established
+
+    flood = []
+    try:
+        for _ in range(300):
+            s = socketmod.socket(socketmod.AF_UNIX, socketmod.SOCK_STREAM)
+            try:
+                s.connect(sock)
+            except OSError:
+                s.close()
+                break
+            flood.append(s)
+        # regardless of how many the server kept, it must still serve
+        for _ in range(20):
+            resp = stub.Allocate(req, timeout=10)
+            assert len(resp.container_responses[0].devices) == 3
+    finally:
+        for s in flood:
+            s.close()
+    ch.close()
+    # and after the flood drains, fresh connections work again
+    ch2 = grpc.insecure_channel(f"unix://{sock}")
+    resp = dp.DevicePluginStub(ch2).Allocate(req, timeout=10)
+    assert len(resp.container_responses[0].devices) == 3
+    ch2.close()
