@@ -13,4 +13,4 @@ reference: internal/pkg/amdgpu/amdgpu.go:21-27, internal/pkg/hwloc/hwloc.go):
   - native/_healthprobe.so gfx950 MFMA/LDS/HBM deep health probe (HIP)
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"

@@ -2,7 +2,7 @@ from setuptools import find_packages, setup
 
 setup(
     name="k8s-device-plugin-amd",
-    version="0.1.0",
+    version="0.2.0",
     description=(
         "MI355X-native Kubernetes device plugin and node labeller for "
         "AMD Instinct GPUs"
