@@ -82,6 +82,9 @@ def device_plugin_main(argv=None) -> int:
     args = ap.parse_args(argv)
     _setup_logging(args.verbose)
     log = logging.getLogger("amd-device-plugin")
+    # startup banner (reference: main.go:94-121 logs name+version lines)
+    log.info("AMD GPU device plugin for Kubernetes (MI355X-native) v%s",
+             __version__)
 
     from .health import HeartbeatTicker
     from .plugin import (
