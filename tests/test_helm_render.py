@@ -1,0 +1,170 @@
+"""Render the Helm templates with a Go-template-subset interpreter and
+YAML-parse the output — catches structural template bugs (bad indent,
+unbalanced if/end, broken multi-doc splits) that string checks miss.
+The subset covers exactly what this chart uses: {{ .Values.* }} /
+{{ .Release.* }} / {{ .Chart.* }} lookups, {{- if }} / {{- else }} /
+{{- with }} / {{- range }} / {{- end }} blocks, and the
+`toYaml X | indent N` and `X | quote` pipes."""
+
+import os
+import re
+
+import pytest
+import yaml
+
+CHART = os.path.join(os.path.dirname(__file__), "..", "deploy", "helm", "amd-gpu")
+
+
+def _lookup(ctx, dotted):
+    cur = ctx
+    for part in dotted.strip().lstrip(".").split("."):
+        if isinstance(cur, dict) and part in cur:
+            cur = cur[part]
+        else:
+            return None
+    return cur
+
+
+def _to_yaml_indented(val, indent):
+    dumped = yaml.safe_dump(val, default_flow_style=False).rstrip()
+    pad = " " * indent
+    return "\n".join(pad + l for l in dumped.splitlines())
+
+
+def render(text, ctx):
+    out_lines = []
+    stack = [True]       # emit-state per if/with/range block
+    dot_stack = [None]   # `with`/`range` rebind the dot
+    range_items = [None]
+
+    def cur_dot():
+        for d in reversed(dot_stack):
+            if d is not None:
+                return d
+        return None
+
+    for line in text.splitlines():
+        m = re.match(r"\s*\{\{-? (if|with|range) ([^}]+?) -?\}\}\s*$", line)
+        if m:
+            kind, expr = m.group(1), m.group(2).strip()
+            val = _lookup(ctx, expr)
+            emit = bool(val) and all(stack)
+            stack.append(emit)
+            if kind == "if":
+                dot_stack.append(None)
+                range_items.append(None)
+            elif kind == "with":
+                dot_stack.append(val)
+                range_items.append(None)
+            else:  # range: emit body once per item with dot = item
+                dot_stack.append(None)
+                range_items.append((list(val) if val else [], []))
+            continue
+        if re.match(r"\s*\{\{-? else -?\}\}\s*$", line):
+            top = stack.pop()
+            stack.append((not top) and all(stack))
+            continue
+        if re.match(r"\s*\{\{-? end -?\}\}\s*$", line):
+            ritems = range_items.pop()
+            emit = stack.pop()
+            dot_stack.pop()
+            if ritems is not None and emit:
+                items, body = ritems
+                for item in items:
+                    for bl in body:
+                        out_lines.append(_expand(bl, ctx, item))
+            continue
+        if range_items[-1] is not None:
+            range_items[-1][1].append(line)
+            continue
+        if not all(stack):
+            continue
+        out_lines.append(_expand(line, ctx, cur_dot()))
+    return "\n".join(out_lines)
+
+
+def _expand(line, ctx, dot):
+    # whole-line toYaml pipes: `{{ toYaml <expr> | indent N }}`
+    m = re.match(r"^\{\{ toYaml (\S+) \| indent (\d+) \}\}$", line.strip())
+    if m:
+        expr = m.group(1)
+        val = dot if expr == "." else _lookup(ctx, expr)
+        assert val is not None, f"unresolved {expr}"
+        return _to_yaml_indented(val, int(m.group(2)))
+
+    def sub(mm):
+        inner = mm.group(1).strip()
+        quote = False
+        if inner.endswith("| quote"):
+            inner = inner[: -len("| quote")].strip()
+            quote = True
+        # image helpers from _helpers.tpl: repo:tag-or-appversion
+        im = re.match(r'include "amd-gpu\.(dp|labeller)-image" \.$', inner)
+        if im:
+            part = "dp" if im.group(1) == "dp" else "labeller"
+            img = ctx["Values"][part]["image"]
+            tag = img.get("tag") or ctx["Chart"].get("AppVersion", "latest")
+            return f"{img['repository']}:{tag}"
+        # `a | default b` pipe
+        dm = re.match(r"([.\w]+) \| default ([.\w]+)$", inner)
+        if dm:
+            val = _lookup(ctx, dm.group(1)) or _lookup(ctx, dm.group(2))
+            assert val is not None, f"unresolved default in {inner}"
+            return str(val)
+        assert "|" not in inner, f"unsupported pipe: {mm.group(0)}"
+        val = dot if inner == "." else _lookup(ctx, inner)
+        assert val is not None, f"unresolved template path {inner}"
+        s = str(val)
+        return f'"{s}"' if quote else s
+
+    return re.sub(r"\{\{-? ([^}]+?) -?\}\}", sub, line)
+
+
+@pytest.fixture(scope="module")
+def ctx():
+    values = yaml.safe_load(open(os.path.join(CHART, "values.yaml")))
+    chart = yaml.safe_load(open(os.path.join(CHART, "Chart.yaml")))
+    # enable everything so every block renders
+    values.setdefault("labeller", {})["enabled"] = True
+    values.setdefault("nfd", {})["enabled"] = True
+    return {
+        "Values": values,
+        "Release": {"Name": "amd-gpu", "Namespace": "kube-system"},
+        "Chart": {"Name": chart["name"], "Version": chart["version"],
+                  "AppVersion": chart.get("appVersion", "latest")},
+    }
+
+
+@pytest.mark.parametrize("tpl", [
+    "device-plugin.yaml", "labeller.yaml", "rbac.yaml", "serviceaccount.yaml",
+])
+def test_template_renders_to_valid_yaml(ctx, tpl):
+    text = open(os.path.join(CHART, "templates", tpl)).read()
+    rendered = render(text, ctx)
+    docs = [d for d in yaml.safe_load_all(rendered) if d]
+    assert docs, f"{tpl} rendered to nothing"
+    for d in docs:
+        assert "kind" in d and "apiVersion" in d, (tpl, d)
+        assert d.get("metadata", {}).get("name"), (tpl, d)
+
+
+def test_rendered_daemonsets_reference_valid_flags(ctx):
+    """Same flag check the raw DaemonSets get, on the rendered chart."""
+    from k8s_device_plugin_amd.cli import device_plugin_main, labeller_main
+
+    from test_deploy_manifests import _cli_flags
+
+    dp_flags = _cli_flags(device_plugin_main)
+    nl_flags = _cli_flags(labeller_main)
+    for tpl, flags in (("device-plugin.yaml", dp_flags),
+                       ("labeller.yaml", nl_flags)):
+        rendered = render(
+            open(os.path.join(CHART, "templates", tpl)).read(), ctx
+        )
+        for d in yaml.safe_load_all(rendered):
+            if not d or d.get("kind") != "DaemonSet":
+                continue
+            for c in d["spec"]["template"]["spec"].get("containers", []):
+                for arg in c.get("args", []):
+                    if isinstance(arg, str) and arg.startswith("-"):
+                        assert arg in flags, (tpl, arg)
