@@ -168,3 +168,16 @@ def test_rendered_daemonsets_reference_valid_flags(ctx):
                 for arg in c.get("args", []):
                     if isinstance(arg, str) and arg.startswith("-"):
                         assert arg in flags, (tpl, arg)
+
+
+def test_notes_lint(ctx):
+    """NOTES.txt uses inline conditionals the line renderer doesn't
+    model; statically lint instead: balanced if/else/end and every
+    referenced value path resolves."""
+    text = open(os.path.join(CHART, "templates", "NOTES.txt")).read()
+    opens = len(re.findall(r"\{\{-? ?if ", text))
+    ends = len(re.findall(r"\{\{-? ?end ?-?\}\}", text))
+    assert opens == ends, (opens, ends)
+    for path in re.findall(r"\{\{-? ?(?:if )?(\.[.\w]+)", text):
+        assert _lookup(ctx, path) is not None, f"NOTES references {path}"
+    assert "amd.com/gpu" in text
